@@ -1,0 +1,28 @@
+"""mapreduce_amd — an MI355X-native distributed MapReduce framework.
+
+A from-scratch re-design of the capabilities of pakozm/lua-mapreduce
+(reference mapreduce/init.lua:25-33 exports worker, server, utils, tuple,
+persistent_table) for AMD Instinct MI355X nodes:
+
+  * control plane: TCPStore CAS (mapreduce_amd.parallel.coord) instead of
+    MongoDB collections;
+  * host data plane: binary record files over mem/shared storage
+    (mapreduce_amd.fs) instead of GridFS;
+  * GPU data plane: HBM-resident partitioned (key, value) tensors, shuffled
+    with RCCL all-to-all over xGMI, sorted/combined/reduced by hand-written
+    CDNA4 HIP kernels (mapreduce_amd.gpu, mapreduce_amd/ops/hip/);
+  * same user contract: task scripts provide init + taskfn/mapfn/
+    partitionfn/reducefn[/combinerfn/finalfn] and reducer property flags
+    (SURVEY.md §2.3), with optional GPU entry points for the fused path.
+"""
+
+from . import fs, job, persistent_table, server, task, utils, worker  # noqa: F401
+from .persistent_table import PersistentTable  # noqa: F401
+from .runner import run_local  # noqa: F401
+from .server import Server  # noqa: F401
+from .utils import tuple as tuple_mod  # noqa: F401
+from .utils.tuple import tuple_  # noqa: F401
+from .worker import Worker  # noqa: F401
+
+_NAME = "mapreduce_amd"
+_VERSION = "0.1.0"

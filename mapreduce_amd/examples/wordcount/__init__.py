@@ -1,0 +1,65 @@
+"""WordCount task script — the reference example, re-expressed.
+
+Parity with mapreduce/examples/WordCount (init.lua:55-64: one module
+providing all six roles + reducer property flags).  taskfn emits one map job
+per input file (taskfn.lua:6-12); mapfn streams lines and emits (word, 1)
+(mapfn.lua:3-9); partitionfn is FNV-1a mod NUM_REDUCERS
+(partitionfn.lua:2-16, multiply-before-xor variant); reducefn sums
+(reducefn.lua:1-15) and is declared associative+commutative+idempotent,
+enabling the combiner and the skip-singleton fast path (job.lua:264-274).
+
+init_args: {"files": [paths...], "out": optional path to write "count word"
+lines, "nred": partitions (default 15)}.
+"""
+
+from __future__ import annotations
+
+from mapreduce_amd.utils.tuple import fnv1a32
+
+_CFG = {"files": [], "out": None, "nred": 15}
+RESULTS = {}  # finalfn drops results here when no "out" path is given
+
+
+def init(arg):
+    if arg:
+        _CFG.update(arg)
+
+
+def taskfn(emit):
+    for i, path in enumerate(_CFG["files"]):
+        emit(i + 1, path)
+
+
+def mapfn(key, value, emit):
+    with open(value, "r", encoding="utf-8", errors="surrogateescape") as fh:
+        for line in fh:
+            for w in line.split():
+                emit(w, 1)
+
+
+def partitionfn(key):
+    return fnv1a32(key) % _CFG["nred"]
+
+
+def reducefn(key, values, emit):
+    emit(sum(values))
+
+
+combinerfn = reducefn
+
+associative_reducer = True
+commutative_reducer = True
+idempotent_reducer = True
+
+
+def finalfn(pairs):
+    RESULTS.clear()
+    out = _CFG.get("out")
+    if out:
+        with open(out, "w", encoding="utf-8", errors="surrogateescape") as fh:
+            for key, values in pairs:
+                fh.write(f"{values[0]}\t{key}\n")
+    else:
+        for key, values in pairs:
+            RESULTS[key] = values[0]
+    return True  # remove result files (finalfn.lua:1-9)

@@ -1,0 +1,284 @@
+"""Map/Reduce job execution — the worker-side data path (host tier).
+
+Parity with mapreduce/job.lua: builds the map or reduce closure for one
+claimed job document, implements emit with the inline combiner
+(threshold MAX_MAP_RESULT, job.lua:92-96), sorts + partitions + spills map
+output (:154-228), k-way-merges + reduces + writes results (:230-296), and
+marks BROKEN on crash (:322-342).
+
+This module is the GENERAL tier: arbitrary Python user functions, record
+streams on the host.  When a task module declares GPU entry points
+(mapfn_gpu / reduce is a declared-associative+commutative builtin), the GPU
+engine (mapreduce_amd.gpu.engine) takes the whole hot path instead —
+mirroring the reference's own fast-path split on reducer property flags
+(job.lua:104-106, 264-274).
+"""
+
+from __future__ import annotations
+
+import importlib
+import re
+import time
+import traceback
+from typing import Any, Callable, Dict, List, Optional
+
+from . import fs as fsmod
+from .parallel.coord import Coordinator
+from .task import Task
+from .utils import (MAX_MAP_RESULT, STATUS, gettime, keys_sorted,
+                    merge_iterator)
+from .utils.tuple import tuple_
+
+
+# ---------------------------------------------------------------------------
+# User task-module loading (job.lua job_get_func :66-115)
+# ---------------------------------------------------------------------------
+
+_module_cache: Dict[Any, Any] = {}
+_inited: set = set()
+
+# objects (dicts/closures) registered in-process so the task document stays
+# JSON — only resolvable inside the same process (LocalCoordinator runs);
+# multi-process runs must use importable module names, like the reference's
+# Lua module-name contract (server.lua:427-433).
+_local_registry: Dict[str, Any] = {}
+
+
+def spec_of(obj: Any) -> Any:
+    """Turn a task-module object into a JSON-safe spec."""
+    if obj is None or isinstance(obj, str):
+        return obj
+    import types
+
+    if isinstance(obj, types.ModuleType):
+        return obj.__name__
+    token = f"@local:{id(obj)}"
+    _local_registry[token] = obj
+    return token
+
+
+def load_module(spec: Any):
+    """Load a task-script module: a dotted module name (like the reference's
+    Lua module names, server.lua:427-433), a dict, or any object with the
+    role attributes.  Modules are cached per process; init(init_args) runs
+    once per module per process (job.lua:70-80)."""
+    if spec is None or spec == "nil":
+        return None
+    if isinstance(spec, str):
+        if spec.startswith("@local:"):
+            obj = _local_registry.get(spec)
+            if obj is None:
+                raise RuntimeError(
+                    f"{spec} is an in-process task object but this worker is "
+                    "a different process; pass importable module names for "
+                    "multi-process runs (server.lua:427-433 contract)")
+            return obj
+        if spec in _module_cache:
+            return _module_cache[spec]
+        mod = importlib.import_module(spec)
+        # a module may expose the contract via a get_task()/TASK factory
+        obj = getattr(mod, "TASK", mod)
+        _module_cache[spec] = obj
+        return obj
+    return spec
+
+
+def _get(obj: Any, name: str):
+    if obj is None:
+        return None
+    if isinstance(obj, dict):
+        return obj.get(name)
+    return getattr(obj, name, None)
+
+
+class FnSet:
+    """Resolved user functions + reducer property flags (§2.3 contract)."""
+
+    ROLES = ("taskfn", "mapfn", "partitionfn", "reducefn",
+             "combinerfn", "finalfn")
+
+    def __init__(self, fns: Dict[str, Any], init_args=None):
+        self.modules: Dict[str, Any] = {}
+        self.init_args = init_args
+        for role in self.ROLES:
+            m = load_module(fns.get(role))
+            self.modules[role] = m
+            if m is not None:
+                key = id(m)
+                if key not in _inited:
+                    init = _get(m, "init")
+                    if callable(init):
+                        init(init_args)
+                    _inited.add(key)
+
+        def fn(role):
+            m = self.modules[role]
+            f = _get(m, role)
+            return f if callable(f) else None
+
+        self.taskfn = fn("taskfn")
+        self.mapfn = fn("mapfn")
+        self.partitionfn = fn("partitionfn")
+        self.reducefn = fn("reducefn")
+        self.combinerfn = fn("combinerfn")
+        self.finalfn = fn("finalfn")
+        rmod = self.modules["reducefn"]
+        self.associative = bool(_get(rmod, "associative_reducer"))
+        self.commutative = bool(_get(rmod, "commutative_reducer"))
+        self.idempotent = bool(_get(rmod, "idempotent_reducer"))
+        # GPU tier hooks (optional; see mapreduce_amd.gpu.engine)
+        self.mapfn_gpu = _get(self.modules["mapfn"], "mapfn_gpu")
+        self.reducefn_gpu = _get(rmod, "reducefn_gpu")
+
+    @property
+    def fast_path(self) -> bool:
+        """Skip-singleton fast path allowed (job.lua:264-274)."""
+        return self.associative and self.commutative and self.idempotent
+
+
+def _apply_combiner(combiner: Callable, key: Any, values: list) -> list:
+    out: List[Any] = []
+    combiner(key, values, out.append)
+    return out
+
+
+def _intern(k: Any) -> Any:
+    return tuple_(*k) if isinstance(k, (tuple, list)) else k
+
+
+# ---------------------------------------------------------------------------
+# Job object
+# ---------------------------------------------------------------------------
+
+
+class Job:
+    """One claimed job (job.lua ctor :345-381)."""
+
+    def __init__(self, coord: Coordinator, task: Task, ns: str, doc: dict,
+                 fns: FnSet, storage: str, path: str = ""):
+        self.coord = coord
+        self.task = task
+        self.ns = ns
+        self.doc = doc
+        self.fns = fns
+        self.fs = fsmod.router(storage, path)
+        self.kind = "map" if ns == Task.MAP_JOBS else "reduce"
+
+    # -- status transitions (job.lua:117-152) ------------------------------
+    def _update(self, **fields) -> None:
+        key = f"{self.ns}/{self.doc['_id']}"
+        for _ in range(64):
+            cur, raw = self.coord.get_doc(key)
+            if cur is None:
+                return
+            if cur["status"] == STATUS.FAILED:
+                return  # server gave up on this job; drop our update
+            new = dict(cur)
+            new.update(fields)
+            if self.coord.cas_doc(key, raw, new):
+                self.doc = new
+                return
+
+    def mark_as_finished(self, cpu_time: float) -> None:
+        self._update(status=STATUS.FINISHED, finished_time=gettime(),
+                     cpu_time=cpu_time)
+
+    def mark_as_written(self, cpu_time: float) -> None:
+        now = gettime()
+        started = self.doc.get("started_time") or now
+        self._update(status=STATUS.WRITTEN, written_time=now,
+                     cpu_time=cpu_time, real_time=now - started)
+
+    def mark_as_broken(self) -> None:
+        """Crash path: BROKEN + $inc repetitions (job.lua:322-342)."""
+        key = f"{self.ns}/{self.doc['_id']}"
+        for _ in range(64):
+            cur, raw = self.coord.get_doc(key)
+            if cur is None or cur["status"] in (STATUS.FAILED, STATUS.WRITTEN):
+                return
+            new = dict(cur)
+            new["status"] = STATUS.BROKEN
+            new["repetitions"] = cur["repetitions"] + 1
+            if self.coord.cas_doc(key, raw, new):
+                self.doc = new
+                return
+
+    # -- execution ---------------------------------------------------------
+    def execute(self) -> None:
+        if self.kind == "map":
+            self._execute_map()
+        else:
+            self._execute_reduce()
+
+    def _execute_map(self) -> None:
+        """job_prepare_map (job.lua:154-228): run mapfn over the split,
+        sort keys (K1), combine (K5), partition (K2), spill one record file
+        per touched partition named map_results.P<p>.M<id> (C5)."""
+        fns = self.fns
+        t0 = time.process_time()
+        result: Dict[Any, list] = {}
+        combiner = fns.combinerfn
+
+        def emit(k, v):
+            k = _intern(k)
+            vs = result.get(k)
+            if vs is None:
+                result[k] = [v]
+                return
+            vs.append(v)
+            if combiner is not None and len(vs) > MAX_MAP_RESULT:
+                result[k] = _apply_combiner(combiner, k, vs)
+
+        job_key = self.doc["_id"]
+        job_value = self.doc["job"]
+        fns.mapfn(job_key, job_value, emit)
+        self.mark_as_finished(time.process_time() - t0)
+
+        t1 = time.process_time()
+        builders: Dict[int, Any] = {}
+        for key in keys_sorted(result):
+            values = result[key]
+            if combiner is not None and len(values) > 1:
+                values = _apply_combiner(combiner, key, values)
+            part = fns.partitionfn(key)
+            if not isinstance(part, int):
+                raise TypeError(
+                    f"partitionfn must return an integer, got {part!r}")
+            b = builders.get(part)
+            if b is None:
+                b = self.fs.builder(f"map_results.P{part}.M{job_key}")
+                builders[part] = b
+            b.append(key, values)
+        for part, b in builders.items():
+            name = f"map_results.P{part}.M{job_key}"
+            self.fs.remove(name)  # idempotent re-execution (job.lua:219)
+            b.build()
+        self.mark_as_written(time.process_time() - t1)
+
+    def _execute_reduce(self) -> None:
+        """job_prepare_reduce (job.lua:230-296): k-way merge the M mapper
+        files of one partition (C6/K4), reduce each key's merged value list
+        (skip-singleton fast path when the reducer is declared
+        assoc+comm+idem, job.lua:264-274), write result.P<p> (C7), delete
+        consumed inputs."""
+        fns = self.fns
+        t0 = time.process_time()
+        value = self.doc["job"]
+        file = value["file"]
+        result_name = value["result"]
+        files = self.fs.list(rf"^{re.escape(file)}\.M.*$")
+        builder = self.fs.builder(result_name)
+        fast = fns.fast_path
+        reducefn = fns.reducefn
+        for key, values in merge_iterator([self.fs.records(f) for f in files]):
+            if fast and len(values) == 1:
+                out = values
+            else:
+                out = []
+                reducefn(key, values, out.append)
+            builder.append(key, out)
+        self.fs.remove(result_name)
+        builder.build()
+        self.mark_as_written(time.process_time() - t0)
+        for f in files:
+            self.fs.remove(f)

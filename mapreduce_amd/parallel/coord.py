@@ -1,0 +1,202 @@
+"""Control plane: the MongoDB replacement.
+
+The reference routes ALL coordination through MongoDB collections — job
+queues, atomic claims, status transitions, the task singleton, the error
+channel and persistent_table (cnn.lua,
+task.lua:258-343).  The MI355X-native equivalent keeps the data plane in HBM
+/ RCCL and needs only a tiny, low-latency host-side KV with CAS for the
+control plane.  We use torch.distributed.TCPStore:
+
+  * Mongo single-doc atomic update  ->  TCPStore.compare_set (CAS on bytes)
+  * collection insert / find        ->  set / get on namespaced keys
+  * $inc counters                   ->  TCPStore.add
+  * server+workers rendezvous       ->  one master store, elastic clients
+    (workers can join/leave at any time, like processes pointed at the same
+    mongod — README.md:13-16)
+
+Two implementations share one interface: ``LocalCoordinator`` (in-process,
+threads) and ``StoreCoordinator`` (TCPStore / any torch Store, multi-process,
+multi-host).  Documents are JSON bytes; CAS compares the exact bytes read.
+"""
+
+from __future__ import annotations
+
+import json
+import threading
+import time
+from typing import Any, Dict, List, Optional, Tuple
+
+
+def _enc(doc: dict) -> bytes:
+    return json.dumps(doc, sort_keys=True, separators=(",", ":")).encode()
+
+
+def _dec(raw: bytes) -> dict:
+    return json.loads(raw.decode())
+
+
+class Coordinator:
+    """Interface; see LocalCoordinator/StoreCoordinator."""
+
+    # --- documents -------------------------------------------------------
+    def set_doc(self, key: str, doc: dict) -> None:
+        raise NotImplementedError
+
+    def get_doc(self, key: str) -> Tuple[Optional[dict], Optional[bytes]]:
+        """Returns (doc, raw_bytes) — raw_bytes is the CAS token."""
+        raise NotImplementedError
+
+    def cas_doc(self, key: str, expected_raw: Optional[bytes], doc: dict) -> bool:
+        """Atomic compare-and-swap.  expected_raw=None means create-if-absent.
+
+        This is the claim primitive replacing Mongo's update-then-find_one
+        (task.lua:301-309): a worker that loses the race simply fails the CAS
+        — no release-if-lost dance needed (task.lua:331-341 becomes moot).
+        """
+        raise NotImplementedError
+
+    def delete_doc(self, key: str) -> None:
+        raise NotImplementedError
+
+    # --- namespaces (job collections) ------------------------------------
+    def set_ids(self, ns: str, ids: List[str]) -> None:
+        self.set_doc(f"{ns}/ids", {"ids": ids})
+
+    def get_ids(self, ns: str) -> List[str]:
+        doc, _ = self.get_doc(f"{ns}/ids")
+        return doc["ids"] if doc else []
+
+    def drop_ns(self, ns: str) -> None:
+        for i in self.get_ids(ns):
+            self.delete_doc(f"{ns}/{i}")
+        self.delete_doc(f"{ns}/ids")
+
+    # --- counters ---------------------------------------------------------
+    def add(self, key: str, n: int) -> int:
+        raise NotImplementedError
+
+    # --- error channel (cnn.lua:62-78) ------------------------------------
+    def insert_error(self, who: str, msg: str) -> None:
+        i = self.add("errors.count", 1)
+        self.set_doc(f"errors/{i}", {"who": who, "msg": msg, "t": time.time()})
+
+    def get_errors(self, drained: int) -> Tuple[List[dict], int]:
+        """Return errors with index > drained and the new high-water mark."""
+        n = self.add("errors.count", 0)
+        out = []
+        for i in range(drained + 1, n + 1):
+            doc, _ = self.get_doc(f"errors/{i}")
+            if doc is not None:
+                out.append(doc)
+        return out, n
+
+    def close(self) -> None:
+        pass
+
+
+class LocalCoordinator(Coordinator):
+    """In-process coordinator for single-process (threaded-worker) runs and
+    unit tests.  Same CAS semantics as the store-backed one."""
+
+    def __init__(self):
+        self._lock = threading.Lock()
+        self._kv: Dict[str, bytes] = {}
+        self._ctr: Dict[str, int] = {}
+
+    def set_doc(self, key: str, doc: dict) -> None:
+        with self._lock:
+            self._kv[key] = _enc(doc)
+
+    def get_doc(self, key: str):
+        with self._lock:
+            raw = self._kv.get(key)
+        return (None, None) if raw is None else (_dec(raw), raw)
+
+    def cas_doc(self, key: str, expected_raw: Optional[bytes], doc: dict) -> bool:
+        new = _enc(doc)
+        with self._lock:
+            cur = self._kv.get(key)
+            if cur != expected_raw:
+                return False
+            self._kv[key] = new
+            return True
+
+    def delete_doc(self, key: str) -> None:
+        with self._lock:
+            self._kv.pop(key, None)
+
+    def add(self, key: str, n: int) -> int:
+        with self._lock:
+            self._ctr[key] = self._ctr.get(key, 0) + n
+            return self._ctr[key]
+
+
+class StoreCoordinator(Coordinator):
+    """TCPStore-backed coordinator: multi-process / multi-host control plane.
+
+    connection string: "tcp://host:port" — the server passes listen=True and
+    hosts the master store; any number of worker processes connect as
+    clients, join, and leave at will (elastic, like the reference's
+    DB-mediated workers).  Any torch.distributed Store (e.g. the PrefixStore
+    of an existing torchrun process group) can be injected via ``store=``.
+    """
+
+    # TCPStore.compare_set with expected=="" creates the key if absent and
+    # returns the desired value; otherwise returns the current value.
+
+    def __init__(self, cnn_string: str = "", db: str = "mr", listen: bool = False,
+                 store=None, timeout_s: float = 300.0):
+        self.db = db
+        if store is not None:
+            self._store = store
+        else:
+            assert cnn_string.startswith("tcp://"), cnn_string
+            host, port = cnn_string[len("tcp://"):].rsplit(":", 1)
+            import datetime
+            from torch.distributed import TCPStore
+
+            self._store = TCPStore(
+                host, int(port), None, listen,
+                timeout=datetime.timedelta(seconds=timeout_s),
+                use_libuv=True, wait_for_workers=False,
+            )
+
+    def _k(self, key: str) -> str:
+        return f"{self.db}/{key}"
+
+    def set_doc(self, key: str, doc: dict) -> None:
+        self._store.set(self._k(key), _enc(doc))
+
+    def get_doc(self, key: str):
+        if not self._store.check([self._k(key)]):
+            return (None, None)
+        raw = bytes(self._store.get(self._k(key)))
+        return _dec(raw), raw
+
+    def cas_doc(self, key: str, expected_raw: Optional[bytes], doc: dict) -> bool:
+        new = _enc(doc)
+        exp = b"" if expected_raw is None else expected_raw
+        got = bytes(self._store.compare_set(self._k(key), exp, new))
+        return got == new
+
+    def delete_doc(self, key: str) -> None:
+        try:
+            self._store.delete_key(self._k(key))
+        except Exception:
+            pass
+
+    def add(self, key: str, n: int) -> int:
+        return self._store.add(self._k(key), n)
+
+
+def connect(cnn_string: str, db: str = "mr", listen: bool = False,
+            store=None) -> Coordinator:
+    """Build a coordinator from a connection string (utils.lua:62-69 analogue).
+
+    "local" -> in-process; "tcp://host:port" -> TCPStore control plane.
+    """
+    if store is not None:
+        return StoreCoordinator(db=db, store=store)
+    if cnn_string in ("local", "", None):
+        return LocalCoordinator()
+    return StoreCoordinator(cnn_string, db=db, listen=listen)

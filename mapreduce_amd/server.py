@@ -1,0 +1,287 @@
+"""Server (driver/orchestrator).
+
+Parity with mapreduce/server.lua: configures a task from user function
+modules, inserts map jobs, polls until all maps are WRITTEN, creates reduce
+jobs from the shuffle files, polls reduces, computes per-phase statistics,
+runs finalfn, supports finalfn->"loop" for iterative MapReduce, and restores
+half-finished tasks on restart (loop :466-611, restore :470-504,
+prepare_map :249-276, prepare_reduce :279-329, final :348-413, stats
+:540-601).
+
+Differences by design (MI355X-native):
+  * control plane = TCPStore CAS (mapreduce_amd.parallel.coord), not Mongo;
+  * stats are reduced in-process from the job documents instead of Mongo's
+    server-side JS map-reduce (server.lua:155-183);
+  * an optional heartbeat timeout requeues jobs of dead workers — a
+    liveness gap in the reference.
+"""
+
+from __future__ import annotations
+
+import re
+import sys
+import time
+from typing import Any, Dict, List, Optional
+
+from . import fs as fsmod
+from .job import FnSet
+from .parallel.coord import Coordinator, connect
+from .task import Task, make_job
+from .utils import (DEFAULT_SLEEP, MAX_TASKFN_VALUE_SIZE, STATUS,
+                    TASK_STATUS, assert_check, gettime)
+
+
+class Server:
+    def __init__(self, cnn_string: str = "local", db: str = "mr",
+                 coord: Optional[Coordinator] = None, listen: bool = True):
+        self.coord = coord or connect(cnn_string, db, listen=listen)
+        self.task = Task(self.coord)
+        self.params: Optional[dict] = None
+        self.finished = False
+        self.iteration = 1
+        self._errors_drained = 0
+        self.stats: Dict[str, Any] = {}
+        self.poll_interval = DEFAULT_SLEEP
+        self.heartbeat_timeout: Optional[float] = None
+        self.verbose = True
+
+    # ------------------------------------------------------------ configure
+    def configure(self, params: dict) -> "Server":
+        """Validate + normalize config (server.lua:419-462).
+
+        params: fns = {taskfn, mapfn, partitionfn, reducefn[, combinerfn,
+        finalfn]} (module names or objects; one object may provide all
+        roles), storage = "mem|shared[:path]", result_ns, init_args.
+        """
+        fns = dict(params.get("fns") or {})
+        for role in ("taskfn", "mapfn", "partitionfn", "reducefn"):
+            if not fns.get(role):
+                raise ValueError(f"needs a {role} module (server.lua:427-428)")
+        fns.setdefault("combinerfn", None)
+        fns.setdefault("finalfn", None)
+        from .job import spec_of
+
+        fns = {role: spec_of(m) for role, m in fns.items()}
+        storage = params.get("storage") or "shared"
+        self.params = {
+            "fns": fns,
+            "storage": storage,
+            "path": params.get("path", ""),
+            "result_ns": params.get("result_ns", "result"),
+            "init_args": params.get("init_args"),
+        }
+        self.poll_interval = params.get("poll_interval", DEFAULT_SLEEP)
+        self.heartbeat_timeout = params.get("heartbeat_timeout")
+        self.verbose = params.get("verbose", True)
+        self.fns = FnSet(fns, self.params["init_args"])
+        self.fs = fsmod.router(storage, self.params["path"])
+        return self
+
+    def _log(self, msg: str) -> None:
+        if self.verbose:
+            print(f"# {msg}", file=sys.stderr, flush=True)
+
+    # ------------------------------------------------------------ map phase
+    def _prepare_map(self) -> None:
+        """server_prepare_map (server.lua:249-276): run taskfn(emit), check
+        key uniqueness and value size, insert map job docs, set phase MAP."""
+        done = self.task.written_ids(Task.MAP_JOBS)
+        self.task.remove_pending(Task.MAP_JOBS)
+        jobs: List[dict] = []
+        seen = set(done)
+        order: List[str] = list(self.task.coord.get_ids(Task.MAP_JOBS))
+
+        def emit(key, value):
+            k = str(key)
+            if k in seen:
+                if k in done:
+                    return  # restored: already WRITTEN, never redo
+                raise ValueError(f"duplicate taskfn key {key!r} "
+                                 "(server.lua:258-261)")
+            seen.add(k)
+            assert_check(value)
+            import pickle
+            if len(pickle.dumps(value)) > MAX_TASKFN_VALUE_SIZE:
+                raise ValueError(
+                    f"taskfn value for key {key!r} exceeds "
+                    f"{MAX_TASKFN_VALUE_SIZE} bytes (server.lua:262-267)")
+            jobs.append(make_job(k, value))
+            order.append(k)
+
+        self.fns.taskfn(emit)
+        for j in jobs:
+            self.task.coord.set_doc(f"{Task.MAP_JOBS}/{j['_id']}", j)
+        self.task.coord.set_ids(Task.MAP_JOBS, order)
+        self.task.set_task_status(TASK_STATUS.MAP)
+        self._log(f"map phase: {len(jobs)} jobs "
+                  f"({len(done)} restored as WRITTEN)")
+
+    # --------------------------------------------------------- reduce phase
+    def _prepare_reduce(self) -> None:
+        """server_prepare_reduce (server.lua:279-329): scan the shuffle
+        namespace for map_results.P<p>.M<m> spills, build one reduce job
+        per partition, set phase REDUCE."""
+        done = self.task.written_ids(Task.RED_JOBS)
+        self.task.remove_pending(Task.RED_JOBS)
+        names = self.fs.list(r"^map_results\.P\d+\.M.*$")
+        parts: Dict[int, int] = {}
+        rx = re.compile(r"^map_results\.P(\d+)\.M(.+)$")
+        for n in names:
+            m = rx.match(n)
+            if m:
+                p = int(m.group(1))
+                parts[p] = parts.get(p, 0) + 1
+        jobs = []
+        order = list(self.task.coord.get_ids(Task.RED_JOBS))
+        for p in sorted(parts):
+            jid = str(p)
+            if jid in done:
+                continue
+            jobs.append(make_job(jid, {
+                "part": p,
+                "file": f"map_results.P{p}",
+                "result": f"{self.params['result_ns']}.P{p}",
+                "mappers": parts[p],
+            }))
+            order.append(jid)
+        for j in jobs:
+            self.task.coord.set_doc(f"{Task.RED_JOBS}/{j['_id']}", j)
+        self.task.coord.set_ids(Task.RED_JOBS, order)
+        self.task.set_task_status(TASK_STATUS.REDUCE)
+        nfiles = sum(parts.values())
+        self._log(f"reduce phase: {len(jobs)} partitions over "
+                  f"{nfiles} shuffle files")
+
+    # ---------------------------------------------------------- poll engine
+    def _poll_until_done(self, ns: str, phase: str) -> None:
+        """Progress poller (make_task_coroutine_wrap, server.lua:186-234):
+        promote exhausted BROKEN jobs to FAILED, requeue stale RUNNING jobs
+        (heartbeat), print % progress, drain the error channel."""
+        last_pct = -1
+        while True:
+            self.task.promote_broken(ns)
+            if self.heartbeat_timeout:
+                n = self.task.requeue_stale(ns, self.heartbeat_timeout)
+                if n:
+                    self._log(f"requeued {n} stale {phase} jobs")
+            errors, self._errors_drained = self.coord.get_errors(
+                self._errors_drained)
+            for e in errors:
+                self._log(f"worker error [{e['who']}]: {e['msg']}")
+            written, failed, total = self.task.count_done(ns)
+            if total:
+                pct = int(100 * (written + failed) / total)
+                if pct != last_pct:
+                    self._log(f"{phase} {pct:3d}% ({written} written, "
+                              f"{failed} failed, {total} total)")
+                    last_pct = pct
+            if written + failed >= total:
+                self.stats[f"{phase}_failed"] = failed
+                return
+            time.sleep(self.poll_interval)
+
+    # --------------------------------------------------------------- stats
+    def _compute_stats(self, ns: str, phase: str) -> None:
+        """Aggregate per-job timing (server.lua compute_sum :177-183,
+        compute_real_time :155-175: cluster time = max(written_time) -
+        min(started_time))."""
+        docs = [d for d in self.task.scan_jobs(ns)
+                if d["status"] == STATUS.WRITTEN]
+        if not docs:
+            return
+        cpu = sum(d["cpu_time"] for d in docs)
+        real = sum(d["real_time"] for d in docs)
+        started = min(d["started_time"] for d in docs if d["started_time"])
+        written = max(d["written_time"] for d in docs if d["written_time"])
+        self.stats[phase] = {
+            "sum_cpu_time": cpu,
+            "sum_real_time": real,
+            "cluster_time": written - started,
+            "jobs": len(docs),
+        }
+
+    def print_stats(self) -> None:
+        """Stats block in the reference's report format (server.lua:557-602)."""
+        for phase in ("map", "reduce"):
+            s = self.stats.get(phase)
+            if not s:
+                continue
+            self._log(f"{phase}: sum(cpu_time) {s['sum_cpu_time']:.2f} s  "
+                      f"sum(real_time) {s['sum_real_time']:.2f} s  "
+                      f"cluster_time {s['cluster_time']:.2f} s  "
+                      f"jobs {s['jobs']}  failed "
+                      f"{self.stats.get(phase + '_failed', 0)}")
+        if "total_time" in self.stats:
+            self._log(f"total server time {self.stats['total_time']:.2f} s")
+
+    # --------------------------------------------------------------- final
+    def _final(self) -> Any:
+        """server_final (server.lua:348-413): stream all result.P<p> files
+        sorted by partition into finalfn as a (key, values) iterator."""
+        rns = self.params["result_ns"]
+        names = self.fs.list(rf"^{re.escape(rns)}\.P\d+$")
+        names.sort(key=lambda n: int(n.rsplit("P", 1)[1]))
+
+        def pair_iterator():
+            for n in names:
+                yield from self.fs.records(n)
+
+        reply = True
+        if self.fns.finalfn is not None:
+            reply = self.fns.finalfn(pair_iterator())
+        if reply is True or reply == "loop":
+            for n in names:
+                self.fs.remove(n)
+        return reply
+
+    # ---------------------------------------------------------------- loop
+    def loop(self) -> None:
+        """Main driver loop (server.lua:466-611)."""
+        assert self.params is not None, "configure() first"
+        t_start = gettime()
+        # restore check (server.lua:470-504)
+        self.task.update()
+        skip_map = False
+        if self.task.exists():
+            st = self.task.status()
+            if st == TASK_STATUS.FINISHED:
+                self.task.drop_all()
+            else:
+                self._log("WARNING: TRYING TO RESTORE A BROKEN TASK "
+                          "(server.lua:479)")
+                self.iteration = self.task.iteration() or 1
+                if st == TASK_STATUS.REDUCE:
+                    skip_map = True
+
+        while not self.finished:
+            if not self.task.exists() or not skip_map:
+                self.task.create_collection(TASK_STATUS.WAIT, self.params,
+                                            self.iteration)
+            if not skip_map:
+                self._prepare_map()
+                self._poll_until_done(Task.MAP_JOBS, "map")
+                self._compute_stats(Task.MAP_JOBS, "map")
+            skip_map = False
+            self._prepare_reduce()
+            self._poll_until_done(Task.RED_JOBS, "reduce")
+            self._compute_stats(Task.RED_JOBS, "reduce")
+            reply = self._final()
+            if reply == "loop":
+                self.iteration += 1
+                self.task.drop_jobs()
+                self._log(f"iterative loop -> iteration {self.iteration}")
+            else:
+                self.finished = True
+        self.task.set_task_status(TASK_STATUS.FINISHED)
+        self.task.drop_jobs()
+        self.stats["total_time"] = gettime() - t_start
+        self.print_stats()
+
+    def drop_all(self) -> None:
+        self.task.update()
+        self.task.drop_all()
+
+
+def new(cnn_string: str = "local", db: str = "mr", **kw) -> Server:
+    """server.new (server.lua:616-624)."""
+    return Server(cnn_string, db, **kw)

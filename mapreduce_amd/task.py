@@ -1,0 +1,256 @@
+"""Shared task state + atomic job claiming.
+
+Parity with mapreduce/task.lua: a singleton "task" document holds the global
+phase (WAIT/MAP/REDUCE/FINISHED), the user function module names, storage and
+iteration counter (:27-58); workers poll it each loop; jobs are claimed
+atomically.  The Mongo update-then-find_one claim (:301-309, with
+release-if-lost :331-341) becomes a plain CAS on the job document — a loser
+fails the CAS and nothing needs releasing.
+
+Map-job -> worker affinity across iterations (:279-293): a worker prefers job
+ids it already executed so rank-resident input buffers are reused
+(SURVEY.md §2.4 "iteration-level reuse"); after MAX_IDLE_COUNT idle polls it
+relaxes to any claimable job.
+"""
+
+from __future__ import annotations
+
+import time
+from typing import Any, Dict, List, Optional, Tuple
+
+from .parallel.coord import Coordinator
+from .utils import (MAX_IDLE_COUNT, MAX_JOB_RETRIES, STATUS, TASK_STATUS,
+                    gettime)
+
+
+def make_job(job_id: str, value: Any) -> dict:
+    """Job document (utils.lua:87-98)."""
+    return {
+        "_id": str(job_id),
+        "job": value,
+        "status": STATUS.WAITING,
+        "worker": None,
+        "tmpname": None,
+        "creation_time": gettime(),
+        "started_time": None,
+        "finished_time": None,
+        "written_time": None,
+        "cpu_time": 0.0,
+        "real_time": 0.0,
+        "repetitions": 0,
+    }
+
+
+class Task:
+    """View over the task singleton + job namespaces (task.lua ctor :345-359:
+    collections <db>.task, <db>.map_jobs, <db>.red_jobs; results namespaces
+    map_results / red_results)."""
+
+    MAP_JOBS = "map_jobs"
+    RED_JOBS = "red_jobs"
+
+    def __init__(self, coord: Coordinator):
+        self.coord = coord
+        self._doc: Optional[dict] = None
+        self._cache_map_ids: set = set()  # executed map ids (affinity cache)
+        self._idle_count = 0
+
+    # ------------------------------------------------------------------ task
+    def create_collection(self, status: str, params: dict, iteration: int) -> None:
+        """Upsert the task singleton (task.lua:96-116)."""
+        doc = {
+            "_id": "unique",
+            "status": status,
+            "iteration": iteration,
+            "fns": params["fns"],
+            "init_args": params.get("init_args"),
+            "storage": params["storage"],
+            "path": params.get("path", ""),
+            "result_ns": params.get("result_ns", "result"),
+        }
+        self.coord.set_doc("task", doc)
+        self._doc = doc
+
+    def update(self) -> None:
+        """Refresh the cached task doc (task.lua:148-160)."""
+        doc, _ = self.coord.get_doc("task")
+        self._doc = doc
+
+    def exists(self) -> bool:
+        return self._doc is not None
+
+    def status(self) -> Optional[str]:
+        return self._doc["status"] if self._doc else None
+
+    def iteration(self) -> int:
+        return self._doc["iteration"] if self._doc else 0
+
+    def fields(self) -> dict:
+        return self._doc or {}
+
+    def set_task_status(self, status: str, **extra) -> None:
+        """Phase transition visible to all workers (task.lua:182-193)."""
+        assert self._doc is not None
+        self._doc["status"] = status
+        self._doc.update(extra)
+        self.coord.set_doc("task", self._doc)
+
+    def set_field(self, key: str, value) -> None:
+        assert self._doc is not None
+        self._doc[key] = value
+        self.coord.set_doc("task", self._doc)
+
+    def finished(self) -> bool:
+        return self.status() in (None, TASK_STATUS.FINISHED, TASK_STATUS.WAIT)
+
+    def get_jobs_ns(self) -> Optional[str]:
+        s = self.status()
+        if s == TASK_STATUS.MAP:
+            return self.MAP_JOBS
+        if s == TASK_STATUS.REDUCE:
+            return self.RED_JOBS
+        return None
+
+    # ------------------------------------------------------------------ jobs
+    def insert_jobs(self, ns: str, jobs: List[dict]) -> None:
+        """Bulk insert of job documents + id index (C1: the scatter of work
+        descriptors; server.lua:271, cnn.lua:80-111 batched inserts)."""
+        for j in jobs:
+            self.coord.set_doc(f"{ns}/{j['_id']}", j)
+        self.coord.set_ids(ns, [j["_id"] for j in jobs])
+
+    def _claimable(self, doc: dict) -> bool:
+        if doc["status"] == STATUS.WAITING:
+            return True
+        return (doc["status"] == STATUS.BROKEN
+                and doc["repetitions"] < MAX_JOB_RETRIES)
+
+    def _try_claim(self, ns: str, job_id: str, worker: str,
+                   tmpname: str) -> Optional[dict]:
+        doc, raw = self.coord.get_doc(f"{ns}/{job_id}")
+        if doc is None or not self._claimable(doc):
+            return None
+        new = dict(doc)
+        new["status"] = STATUS.RUNNING
+        new["worker"] = worker
+        new["tmpname"] = tmpname
+        new["started_time"] = gettime()
+        if self.coord.cas_doc(f"{ns}/{job_id}", raw, new):
+            return new
+        return None
+
+    def take_next_job(self, worker: str, tmpname: str
+                      ) -> Tuple[Optional[str], Optional[dict]]:
+        """Claim one job, honoring iteration affinity (task.lua:258-343).
+
+        Returns (ns, job_doc) or (None, None) when nothing is claimable
+        (sleep path).  Map affinity: on iterations > 1 prefer ids this
+        worker already executed (warm HBM input buffers); relax after
+        MAX_IDLE_COUNT misses (:284-292).
+        """
+        ns = self.get_jobs_ns()
+        if ns is None:
+            return None, None
+        ids = self.coord.get_ids(ns)
+        candidates = ids
+        if (ns == self.MAP_JOBS and self.iteration() > 1
+                and self._cache_map_ids and self._idle_count < MAX_IDLE_COUNT):
+            cached = [i for i in ids if i in self._cache_map_ids]
+            candidates = cached if cached else ids
+        for job_id in candidates:
+            doc = self._try_claim(ns, job_id, worker, tmpname)
+            if doc is not None:
+                self._idle_count = 0
+                if ns == self.MAP_JOBS:
+                    self._cache_map_ids.add(job_id)
+                return ns, doc
+        self._idle_count += 1
+        return None, None
+
+    # ------------------------------------------------- server-side job admin
+    def scan_jobs(self, ns: str) -> List[dict]:
+        out = []
+        for i in self.coord.get_ids(ns):
+            doc, _ = self.coord.get_doc(f"{ns}/{i}")
+            if doc is not None:
+                out.append(doc)
+        return out
+
+    def promote_broken(self, ns: str) -> int:
+        """BROKEN with repetitions >= MAX_JOB_RETRIES -> FAILED
+        (server.lua:192-205).  Returns number promoted."""
+        n = 0
+        for i in self.coord.get_ids(ns):
+            while True:
+                doc, raw = self.coord.get_doc(f"{ns}/{i}")
+                if (doc is None or doc["status"] != STATUS.BROKEN
+                        or doc["repetitions"] < MAX_JOB_RETRIES):
+                    break
+                new = dict(doc)
+                new["status"] = STATUS.FAILED
+                if self.coord.cas_doc(f"{ns}/{i}", raw, new):
+                    n += 1
+                    break
+        return n
+
+    def requeue_stale(self, ns: str, timeout_s: float) -> int:
+        """Liveness repair the reference lacks (SURVEY.md §5 'a dead worker's
+        RUNNING job is not auto-requeued'): RUNNING jobs older than
+        timeout_s with no heartbeat go back to BROKEN (+1 repetition) so
+        another worker can reclaim them."""
+        n = 0
+        now = gettime()
+        for i in self.coord.get_ids(ns):
+            doc, raw = self.coord.get_doc(f"{ns}/{i}")
+            if doc is None or doc["status"] != STATUS.RUNNING:
+                continue
+            hb = doc.get("heartbeat") or doc.get("started_time") or now
+            if now - hb <= timeout_s:
+                continue
+            new = dict(doc)
+            new["status"] = STATUS.BROKEN
+            new["repetitions"] = doc["repetitions"] + 1
+            if self.coord.cas_doc(f"{ns}/{i}", raw, new):
+                n += 1
+        return n
+
+    def count_done(self, ns: str) -> Tuple[int, int, int]:
+        """(written, failed, total) — the progress poll C4
+        (server.lua:207-231)."""
+        written = failed = total = 0
+        for doc in self.scan_jobs(ns):
+            total += 1
+            if doc["status"] == STATUS.WRITTEN:
+                written += 1
+            elif doc["status"] == STATUS.FAILED:
+                failed += 1
+        return written, failed, total
+
+    def remove_pending(self, ns: str) -> None:
+        """Delete every non-WRITTEN job doc so finished work survives a
+        server restart (server.lua:237-245)."""
+        keep = []
+        for i in self.coord.get_ids(ns):
+            doc, _ = self.coord.get_doc(f"{ns}/{i}")
+            if doc is not None and doc["status"] == STATUS.WRITTEN:
+                keep.append(i)
+            else:
+                self.coord.delete_doc(f"{ns}/{i}")
+        self.coord.set_ids(ns, keep)
+
+    def written_ids(self, ns: str) -> set:
+        out = set()
+        for doc in self.scan_jobs(ns):
+            if doc["status"] == STATUS.WRITTEN:
+                out.add(doc["_id"])
+        return out
+
+    def drop_jobs(self) -> None:
+        self.coord.drop_ns(self.MAP_JOBS)
+        self.coord.drop_ns(self.RED_JOBS)
+
+    def drop_all(self) -> None:
+        """server_drop_collections (server.lua:331-345)."""
+        self.drop_jobs()
+        self.coord.delete_doc("task")
+        self._doc = None

@@ -1,0 +1,209 @@
+"""Core constants, status enums, serialization and merge iterators.
+
+MI355X-native re-implementation of the utility layer of pakozm/lua-mapreduce
+(reference: mapreduce/utils.lua).  The reference keeps job/task state machines,
+tuning constants, a text (key,value) wire format and heap-based k-way merge
+iterators; we keep the same semantics with Python/binary equivalents.  The GPU
+data plane (mapreduce_amd.gpu) replaces the text codec with packed int64
+record tensors; this module is the host-side / general-purpose tier.
+
+Reference parity map:
+  STATUS enum            -> utils.lua:33-40
+  TASK_STATUS enum       -> utils.lua:41-46
+  tuning constants       -> utils.lua:27-55
+  serialize/escape       -> utils.lua:100-120 (here: canonical repr + pickle framing)
+  keys_sorted            -> utils.lua:123-128
+  merge_iterator         -> utils.lua:206-271 (heap-based k-way merge w/ value concat)
+"""
+
+from __future__ import annotations
+
+import io
+import math
+import pickle
+import struct
+import time
+from typing import Any, Callable, Iterable, Iterator, List, Tuple
+
+from .heap import Heap
+
+# ---------------------------------------------------------------------------
+# Job status state machine (utils.lua:33-40)
+# ---------------------------------------------------------------------------
+
+
+class STATUS:
+    WAITING = 0   # ready to be claimed
+    RUNNING = 1   # claimed by a worker
+    BROKEN = 2    # worker crashed while executing; reclaimable
+    FINISHED = 3  # computation done, output not yet durable
+    WRITTEN = 4   # output durable (spill written / tensors published)
+    FAILED = 5    # exceeded MAX_JOB_RETRIES; promoted by the server
+
+
+class TASK_STATUS:
+    """Phase of the whole task (utils.lua:41-46)."""
+
+    WAIT = "WAIT"
+    MAP = "MAP"
+    REDUCE = "REDUCE"
+    FINISHED = "FINISHED"
+
+
+# Tuning constants (utils.lua:27-55).  Names kept for parity; values adapted
+# where the MI355X engine has different natural scales.
+DEFAULT_RW_TIMEOUT = 300
+DEFAULT_SLEEP = 0.05          # reference polls at 1 s against Mongo; a local
+                              # TCP store sustains far faster polling
+DEFAULT_MICRO_SLEEP = 0.005
+DEFAULT_HOSTNAME = "unknown"
+DEFAULT_IP = "127.0.0.1"
+DEFAULT_DATE = 0
+DEFAULT_STORAGE = "shared"
+MAX_PENDING_INSERTS = 50000
+MAX_JOB_RETRIES = 3           # utils.lua:36 MAX_JOB_RETRIES
+MAX_WORKER_RETRIES = 3        # worker gives up after 3 distinct failed jobs
+MAX_MAP_RESULT = 5000         # inline-combiner threshold (job.lua:92-96)
+MAX_TASKFN_VALUE_SIZE = 16 * 1024  # serialized taskfn value limit (server.lua:262-267)
+MAX_IT_WO_CGARBAGE = 5000
+MAX_TIME_WO_CGARBAGE = 60
+MAX_IDLE_COUNT = 5            # affinity relax threshold (task.lua:284-292)
+
+GRP_TMP_DIR = "/tmp/grp_tmp_dir"
+
+
+def gettime() -> float:
+    return time.time()
+
+
+# ---------------------------------------------------------------------------
+# Canonical key ordering + serialization
+# ---------------------------------------------------------------------------
+
+_TYPE_RANK = {int: 0, float: 0, bool: 0, str: 1, bytes: 2, tuple: 3}
+
+
+def sort_key(key: Any):
+    """Total order over supported key types (numbers < strings < bytes < tuples).
+
+    The reference sorts spill keys with Lua table.sort (job.lua:194,
+    utils.lua:123-128), which requires homogeneous keys; we additionally rank
+    by type so mixed-type key spaces have a stable global order.
+    """
+    t = type(key)
+    if t is tuple:
+        return (3, tuple(sort_key(k) for k in key))
+    r = _TYPE_RANK.get(t)
+    if r is None:
+        from .tuple import InternedTuple
+
+        if isinstance(key, (int, float)):
+            r = 0
+        elif isinstance(key, str):
+            r = 1
+        elif isinstance(key, bytes):
+            r = 2
+        elif isinstance(key, (tuple, InternedTuple)):
+            # interned composite keys order like plain tuples; the
+            # length-first ordering of InternedTuple itself only applies
+            # when comparing tuples directly (tuple.lua:183-201)
+            return (3, tuple(sort_key(k) for k in key))
+        else:
+            raise TypeError(f"unsupported key type: {t!r}")
+    return (r, key)
+
+
+def keys_sorted(d: dict) -> list:
+    """Sorted list of a dict's keys (utils.lua:123-128)."""
+    return sorted(d.keys(), key=sort_key)
+
+
+def assert_check(value: Any, path: str = "value") -> None:
+    """Validate a taskfn value is plain data (server-side check analogous to
+    the JSON-compatibility validation at utils.lua:313-333)."""
+    if value is None or isinstance(value, (int, float, str, bytes, bool)):
+        return
+    if isinstance(value, (list, tuple)):
+        for i, v in enumerate(value):
+            assert_check(v, f"{path}[{i}]")
+        return
+    if isinstance(value, dict):
+        for k, v in value.items():
+            if not isinstance(k, (int, float, str, bytes, bool)):
+                raise TypeError(f"{path}: unsupported dict key {type(k)!r}")
+            assert_check(v, f"{path}[{k!r}]")
+        return
+    raise TypeError(f"{path}: unsupported value type {type(value)!r}")
+
+
+# Binary record framing: replaces the reference's "return k,{v...}\n" text rows
+# (utils.lua:100-120 writer, utils.lua:222-224 load() parser).  A record is a
+# length-prefixed pickle of (key, values_list); files are streams of records
+# sorted by sort_key(key).
+_LEN = struct.Struct("<I")
+
+
+def write_record(fh, key: Any, values: list) -> None:
+    payload = pickle.dumps((key, values), protocol=pickle.HIGHEST_PROTOCOL)
+    fh.write(_LEN.pack(len(payload)))
+    fh.write(payload)
+
+
+def read_records(fh) -> Iterator[Tuple[Any, list]]:
+    while True:
+        hdr = fh.read(4)
+        if not hdr:
+            return
+        if len(hdr) != 4:
+            raise EOFError("truncated record header")
+        (n,) = _LEN.unpack(hdr)
+        payload = fh.read(n)
+        if len(payload) != n:
+            raise EOFError("truncated record payload")
+        yield pickle.loads(payload)
+
+
+def serialize_records(pairs: Iterable[Tuple[Any, list]]) -> bytes:
+    buf = io.BytesIO()
+    for k, vs in pairs:
+        write_record(buf, k, vs)
+    return buf.getvalue()
+
+
+# ---------------------------------------------------------------------------
+# k-way merge of sorted record streams (utils.lua:206-271 + heap.lua)
+# ---------------------------------------------------------------------------
+
+
+def merge_iterator(iterators: List[Iterator[Tuple[Any, list]]]) -> Iterator[Tuple[Any, list]]:
+    """Merge N iterators of (key, values) records, each sorted by sort_key,
+    concatenating the value lists of equal keys across streams.
+
+    Mirrors utils.merge_iterator (utils.lua:206-271): a binary min-heap holds
+    one head record per stream; each step pops the minimum key, drains every
+    stream whose head has an equal key (appending values), refills, and yields
+    one (key, merged_values) pair.  On the GPU tier this entire merge is
+    replaced by radix sort + segmented reduce.
+    """
+    heap: Heap = Heap(key=lambda item: item[0])
+
+    def refill(idx: int, it: Iterator) -> None:
+        try:
+            k, vs = next(it)
+        except StopIteration:
+            return
+        heap.push((sort_key(k), k, list(vs), idx))
+
+    its = list(iterators)
+    for i, it in enumerate(its):
+        refill(i, it)
+
+    while not heap.empty():
+        sk, k, vs, idx = heap.pop()
+        refill(idx, its[idx])
+        # concatenate equal keys across streams (utils.lua:238-245)
+        while not heap.empty() and heap.top()[0] == sk:
+            _, _, vs2, idx2 = heap.pop()
+            vs.extend(vs2)
+            refill(idx2, its[idx2])
+        yield k, vs

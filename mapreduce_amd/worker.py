@@ -1,0 +1,136 @@
+"""Worker (executor daemon).
+
+Parity with mapreduce/worker.lua: polls the task singleton, atomically claims
+jobs, executes them, idles with exponential backoff (x1.5 capped at
+max_sleep, worker.lua:100-101), and wraps execution in a crash barrier that
+marks the in-flight job BROKEN, reports to the error channel, and gives up
+after MAX_WORKER_RETRIES distinct failed jobs (worker.lua:112-138).
+
+A worker is elastic: any process (or thread) pointed at the same control
+plane joins the pool, exactly like reference workers pointed at the same
+mongod (README.md:13-16).
+"""
+
+from __future__ import annotations
+
+import os
+import socket
+import threading
+import time
+import traceback
+from typing import Optional
+
+from .job import FnSet, Job
+from .parallel.coord import Coordinator, connect
+from .task import Task
+from .utils import DEFAULT_SLEEP, MAX_WORKER_RETRIES, TASK_STATUS
+
+
+class Worker:
+    def __init__(self, cnn_string: str = "local", db: str = "mr",
+                 coord: Optional[Coordinator] = None, name: str = ""):
+        self.coord = coord or connect(cnn_string, db, listen=False)
+        self.task = Task(self.coord)
+        self.name = name or (f"{socket.gethostname()}:{os.getpid()}:"
+                             f"{threading.get_ident()}")
+        self.tmpname = f"w{os.getpid()}_{threading.get_ident()}"
+        self.max_iter = 20
+        self.max_sleep = 20.0
+        self.min_sleep = DEFAULT_SLEEP
+        self.max_tasks = 1
+        self.verbose = False
+        self._fns_cache = {}
+        self._stop = threading.Event()
+
+    def configure(self, params: dict) -> "Worker":
+        """worker:configure{max_iter, max_sleep, max_tasks}
+        (worker.lua:142-148, defaults :160-163)."""
+        self.max_iter = params.get("max_iter", self.max_iter)
+        self.max_sleep = params.get("max_sleep", self.max_sleep)
+        self.max_tasks = params.get("max_tasks", self.max_tasks)
+        self.min_sleep = params.get("min_sleep", self.min_sleep)
+        self.verbose = params.get("verbose", self.verbose)
+        return self
+
+    def stop(self) -> None:
+        self._stop.set()
+
+    def _log(self, msg: str) -> None:
+        if self.verbose:
+            import sys
+            print(f"# worker {self.name}: {msg}", file=sys.stderr, flush=True)
+
+    def _get_fns(self) -> FnSet:
+        # one FnSet per distinct fns config (module load + init once/process)
+        key = repr(sorted((self.task.fields().get("fns") or {}).items()))
+        fns = self._fns_cache.get(key)
+        if fns is None:
+            fns = FnSet(self.task.fields()["fns"],
+                        self.task.fields().get("init_args"))
+            self._fns_cache[key] = fns
+        return fns
+
+    def _worker_execute(self) -> None:
+        """Main loop (worker.lua:42-105)."""
+        it = 0
+        ntasks = 0
+        sleep = self.min_sleep
+        while (it < self.max_iter and ntasks < self.max_tasks
+               and not self._stop.is_set()):
+            it += 1
+            self.task.update()
+            job_done = False
+            if self.task.exists() and not self.task.finished():
+                while not self._stop.is_set():
+                    self.task.update()
+                    if not self.task.exists() or self.task.finished():
+                        break
+                    ns, doc = self.task.take_next_job(self.name, self.tmpname)
+                    if doc is None:
+                        time.sleep(self.min_sleep)
+                        continue
+                    fields = self.task.fields()
+                    job = Job(self.coord, self.task, ns, doc,
+                              self._get_fns(), fields["storage"],
+                              fields.get("path", ""))
+                    self._current_job = job
+                    self._log(f"executing {ns} job {doc['_id']}")
+                    job.execute()
+                    self._current_job = None
+                    job_done = True
+                    sleep = self.min_sleep
+            if job_done:
+                ntasks += 1
+            if ntasks < self.max_tasks and not self._stop.is_set():
+                self._log(f"idle, sleeping {sleep:.2f}s")
+                time.sleep(sleep)
+                sleep = min(sleep * 1.5, self.max_sleep)  # worker.lua:100-101
+
+    def execute(self) -> None:
+        """Crash-barrier wrapper (worker.lua:112-138): on any exception,
+        mark the in-flight job BROKEN, report the traceback on the error
+        channel, sleep, retry; abort after MAX_WORKER_RETRIES distinct
+        failed jobs."""
+        self._current_job: Optional[Job] = None
+        failed_jobs = set()
+        while len(failed_jobs) < MAX_WORKER_RETRIES and not self._stop.is_set():
+            try:
+                self._worker_execute()
+                return
+            except Exception:
+                tb = traceback.format_exc()
+                job = self._current_job
+                if job is not None:
+                    failed_jobs.add((job.ns, job.doc["_id"]))
+                    job.mark_as_broken()
+                    self._current_job = None
+                self.coord.insert_error(self.name, tb)
+                self._log(f"crash barrier: {tb.splitlines()[-1]}")
+                time.sleep(self.min_sleep)
+        self._log("giving up after repeated failures "
+                  "(worker.lua:133-137)")
+
+
+def new(cnn_string: str = "local", db: str = "mr", **kw) -> Worker:
+    """worker.new (worker.lua:142-167)."""
+    return Worker(cnn_string, db, **kw)
