@@ -1,0 +1,68 @@
+"""Generic server CLI (execute_server.lua:25-62 analogue).
+
+    python -m mapreduce_amd.execute_server tcp://HOST:PORT DBNAME \
+        TASKFN MAPFN PARTITIONFN REDUCEFN [FINALFN] [COMBINERFN] \
+        [--storage shared:/path] [--init-args '{"files": [...]}']
+
+Function arguments are importable module names (dot form); "nil" for an
+absent optional role, like the reference CLI.  The server hosts the TCPStore
+master; workers started with execute_worker against the same tcp:// address
+join elastically.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import sys
+import time
+
+
+def normalize(name: str):
+    if not name or name == "nil":
+        return None
+    return name.replace("/", ".").removesuffix(".py")
+
+
+def main(argv=None) -> int:
+    p = argparse.ArgumentParser(description=__doc__)
+    p.add_argument("cnn", help="tcp://host:port or 'local'")
+    p.add_argument("db")
+    p.add_argument("taskfn")
+    p.add_argument("mapfn")
+    p.add_argument("partitionfn")
+    p.add_argument("reducefn")
+    p.add_argument("finalfn", nargs="?", default="nil")
+    p.add_argument("combinerfn", nargs="?", default="nil")
+    p.add_argument("--storage", default="shared")
+    p.add_argument("--result-ns", default="result")
+    p.add_argument("--init-args", default=None,
+                   help="JSON passed to every module's init()")
+    p.add_argument("--heartbeat-timeout", type=float, default=None)
+    p.add_argument("--sleep", type=float, default=1.0,
+                   help="startup grace before loop (execute_server.lua sleep(4))")
+    args = p.parse_args(argv)
+
+    from .server import Server
+
+    srv = Server(args.cnn, args.db).configure({
+        "fns": {
+            "taskfn": normalize(args.taskfn),
+            "mapfn": normalize(args.mapfn),
+            "partitionfn": normalize(args.partitionfn),
+            "reducefn": normalize(args.reducefn),
+            "finalfn": normalize(args.finalfn),
+            "combinerfn": normalize(args.combinerfn),
+        },
+        "storage": args.storage,
+        "result_ns": args.result_ns,
+        "init_args": json.loads(args.init_args) if args.init_args else None,
+        "heartbeat_timeout": args.heartbeat_timeout,
+    })
+    time.sleep(args.sleep)
+    srv.loop()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())
