@@ -1,0 +1,123 @@
+"""Flagship benchmark: Europarl-shape word count on MI355X.
+
+Measures the BASELINE.json headline metric — words/sec, whole-job wall
+clock — on synthetic text of the named shape (49,158,635 words, 197 splits
+per GPU; weak scaling: each rank owns one Europarl-size corpus).
+
+  python bench.py --gpus 1 --steps 20 --warmup 5
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node 8 \
+      --master-addr 127.0.0.1 bench.py --gpus 8 ...
+
+One step = one complete MapReduce job over the corpus: fused
+tokenize+combine kernels, unique extraction + radix sort, RCCL all-to-all
+shuffle, sort+segmented reduce, per-rank sorted results materialized
+(counts ready for the finalfn boundary).  Reference headline to beat:
+49.23 s / ~1.0 M words/s on 4 CPU workers.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import sys
+import time
+
+import torch
+
+BASELINE_WORDS_PER_SEC = 49_158_635 / 49.23  # BASELINE.md README.md:73
+
+
+def main() -> int:
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--words", type=int, default=49_158_635,
+                   help="words per GPU (Europarl v7 English size)")
+    p.add_argument("--splits", type=int, default=197)
+    p.add_argument("--vocab", type=int, default=130_000)
+    p.add_argument("--device", default=None)
+    args = p.parse_args()
+
+    from mapreduce_amd import ops
+    from mapreduce_amd.gpu import dist as dx
+    from mapreduce_amd.gpu.corpus import make_corpus
+    from mapreduce_amd.gpu.wordcount import WordCountJob
+
+    rank, world, device = dx.init_from_env(args.device)
+    if device.type == "cuda":
+        ops.require_gpu_ext()  # HIP kernels are mandatory on GPU
+
+    corpus = make_corpus(device, nwords=args.words, nsplits=args.splits,
+                         vocab_size=args.vocab, seed=1234 + rank)
+    job = WordCountJob(device, vocab_estimate=max(args.vocab, 1 << 12))
+    splits = corpus.splits()
+
+    def sync():
+        if device.type == "cuda":
+            torch.cuda.synchronize(device)
+
+    def one_step():
+        res = job.run(corpus.text, splits)
+        return res
+
+    # warmup (untimed)
+    res = None
+    for _ in range(args.warmup):
+        res = one_step()
+    sync()
+    dx.barrier()
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        res = one_step()
+    sync()
+    dx.barrier()
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks (the slowest rank defines job wall-clock)
+    el_t = torch.tensor([elapsed], dtype=torch.float64,
+                        device=device if device.type == "cuda" else "cpu")
+    if world > 1:
+        import torch.distributed as td
+        td.all_reduce(el_t, op=td.ReduceOp.MAX)
+    elapsed = float(el_t.item())
+
+    total_words_per_step = args.words * world
+    assert res is not None and res.nwords == args.words, \
+        f"word count mismatch: {res.nwords} != {args.words}"
+    value = total_words_per_step * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        out = {
+            "metric": "words/sec",
+            "value": value,
+            "unit": "words/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": value / BASELINE_WORDS_PER_SEC,
+            "dtype": "u8-text/i64-counts",
+            "data": "synthetic (Europarl v7 shape: 49,158,635 words x N "
+                    "GPUs, 197 splits/GPU, Zipf vocab 130k)",
+            "config": {
+                "model": "europarl-wordcount",
+                "global_batch": total_words_per_step,
+                "seq_len": args.words,
+                "parallelism": f"dp{world}",
+                "splits_per_gpu": args.splits,
+                "partitions": world,
+                "partitioner": "hash-mulhi",
+            },
+        }
+        print(json.dumps(out), flush=True)
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,111 @@
+"""torch.distributed helpers + the shuffle exchange primitive.
+
+One process per GPU over RCCL ("nccl" backend IS RCCL on ROCm); CPU test
+runs use gloo.  The all-to-all is the C5/C6 shuffle of SURVEY.md §2.5: with
+hash-sorted keys and mulhi partitioning the send buffer is partition-
+contiguous, so exchange() is a single uneven all_to_all_single per array.
+xGMI is point-to-point (7 links/GPU), so the all-to-all uses all links
+concurrently — exactly the topology RCCL's alltoall maps to."""
+
+from __future__ import annotations
+
+import os
+from typing import List, Optional, Tuple
+
+import torch
+import torch.distributed as dist
+
+
+def init_from_env(device_type: Optional[str] = None):
+    """Initialize the default process group from torchrun env vars.
+    Returns (rank, world, device).  Single-process (no env) -> (0, 1, dev)
+    without initializing."""
+    if device_type is None:
+        device_type = "cuda" if torch.cuda.is_available() else "cpu"
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    if device_type == "cuda":
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+    else:
+        device = torch.device("cpu")
+    if world > 1 and not dist.is_initialized():
+        backend = "nccl" if device_type == "cuda" else "gloo"
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29531")
+        dist.init_process_group(backend=backend, rank=rank, world_size=world)
+    return rank, world, device
+
+
+def world_info(group=None) -> Tuple[int, int]:
+    if dist.is_available() and dist.is_initialized():
+        return dist.get_rank(group), dist.get_world_size(group)
+    return 0, 1
+
+
+def exchange_counts(send_counts: torch.Tensor, group=None) -> torch.Tensor:
+    """recv_counts[i] = send_counts of rank i toward me (C5 size exchange).
+    send_counts: i64[world] on the comm device."""
+    rank, world = world_info(group)
+    if world == 1:
+        return send_counts.clone()
+    mat = [torch.zeros_like(send_counts) for _ in range(world)]
+    dist.all_gather(mat, send_counts.contiguous(), group=group)
+    return torch.stack([mat[i][rank] for i in range(world)])
+
+
+def exchange(data: torch.Tensor, send_counts: List[int],
+             recv_counts: List[int], group=None) -> torch.Tensor:
+    """Uneven all-to-all of a 1-D tensor sliced by send_counts.
+    Falls back to P2P send/recv where the backend lacks alltoall."""
+    rank, world = world_info(group)
+    if world == 1:
+        return data.clone()
+    out = torch.empty(int(sum(recv_counts)), dtype=data.dtype,
+                      device=data.device)
+    try:
+        dist.all_to_all_single(out, data.contiguous(),
+                               output_split_sizes=recv_counts,
+                               input_split_sizes=send_counts, group=group)
+        return out
+    except (RuntimeError, ValueError):
+        pass
+    # P2P fallback (gloo without alltoall): pairwise rounds
+    soff = [0]
+    for c in send_counts:
+        soff.append(soff[-1] + c)
+    roff = [0]
+    for c in recv_counts:
+        roff.append(roff[-1] + c)
+    out[roff[rank]:roff[rank + 1]] = data[soff[rank]:soff[rank + 1]]
+    reqs = []
+    for peer in range(world):
+        if peer == rank:
+            continue
+        if send_counts[peer]:
+            reqs.append(dist.isend(
+                data[soff[peer]:soff[peer + 1]].contiguous(), dst=peer,
+                tag=rank, group=group))
+    for peer in range(world):
+        if peer == rank:
+            continue
+        if recv_counts[peer]:
+            buf = torch.empty(recv_counts[peer], dtype=data.dtype,
+                              device=data.device)
+            dist.recv(buf, src=peer, tag=peer, group=group)
+            out[roff[peer]:roff[peer + 1]] = buf
+    for r in reqs:
+        r.wait()
+    return out
+
+
+def barrier(group=None):
+    if dist.is_available() and dist.is_initialized():
+        dist.barrier(group=group)
+
+
+def allreduce_sum(t: torch.Tensor, group=None) -> torch.Tensor:
+    if dist.is_available() and dist.is_initialized():
+        dist.all_reduce(t, op=dist.ReduceOp.SUM, group=group)
+    return t

@@ -1,0 +1,122 @@
+"""Flagship fused word-count job — the BASELINE.json headline workload.
+
+Per step (= one full MapReduce job over the rank's corpus):
+
+  MAP+COMBINE   tokenize_count kernel per split: words stream from HBM text
+                straight into the per-rank hash table (the reference's
+                mapfn emit + inline combiner job.lua:83-97, fused into one
+                kernel pass; word count accumulates on-device).
+  EXTRACT+SORT  table -> unique (hash, count, exemplar pos); radix sort by
+                hash (K1) — with mulhi partitioning, the sorted array is
+                partition-contiguous.
+  SHUFFLE       RCCL all-to-all of (hash, count, exemplar len/bytes) slices
+                (C5/C6): one collective per array over the 7 xGMI links.
+  REDUCE        sort the received runs + segmented reduce-by-key (K4->K1+K5),
+                keeping the first exemplar per key.
+  RESULT        per-rank sorted (hash, count, word) arrays; to_host()
+                materializes (word -> count) for the finalfn boundary (C8).
+
+The reducer here is the declared associative+commutative+idempotent sum
+(examples/WordCount flags), which is exactly the reference's own fast-path
+precondition (job.lua:264-274)."""
+
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass
+from typing import List, Optional, Tuple
+
+import torch
+
+from .. import ops
+from . import dist as dx
+
+
+@dataclass
+class WordCountResult:
+    keys: torch.Tensor      # i64 (u64 hash bits), sorted
+    counts: torch.Tensor    # i64
+    pos: torch.Tensor       # packed (off<<16|len) into blob_src
+    blob_src: torch.Tensor  # u8 source for exemplar bytes
+    nwords: int             # words processed by this rank this step
+
+    def to_host(self) -> List[Tuple[bytes, int]]:
+        """Materialize (word, count) pairs, sorted by key hash (C8)."""
+        lens, blob = ops.extract_words(self.blob_src, self.pos)
+        raw = bytes(blob.cpu().numpy().tobytes())
+        counts = self.counts.cpu().tolist()
+        out = []
+        off = 0
+        for L, c in zip(lens.cpu().tolist(), counts):
+            out.append((raw[off:off + L], c))
+            off += L
+        return out
+
+
+class WordCountJob:
+    def __init__(self, device, vocab_estimate: int = 1 << 18, group=None):
+        self.device = torch.device(device)
+        self.group = group
+        self.rank, self.world = dx.world_info(group)
+        self.table = ops.make_table(vocab_estimate, self.device)
+        self._nwords = torch.zeros(1, dtype=torch.int64, device=self.device)
+
+    def reset(self, vocab_estimate: int = 1 << 18):
+        self.table = ops.make_table(vocab_estimate, self.device)
+        self._nwords.zero_()
+
+    def run(self, text: torch.Tensor,
+            splits: Optional[List[Tuple[int, int]]] = None) -> WordCountResult:
+        """One full job over this rank's corpus bytes."""
+        self.reset()
+        dev = self.device
+        if splits is None:
+            splits = [(0, int(text.numel()))]
+
+        # ---- MAP + COMBINE (one fused kernel launch per map job/split)
+        for (s, e) in splits:
+            self.table.tokenize_count(text[s:e], s, self._nwords)
+
+        # ---- EXTRACT + SORT
+        uk, uv, up = self.table.extract()
+        sk, sv, sp = ops.sort_by_key(uk, uv, up)
+
+        if self.world > 1:
+            # ---- SHUFFLE (C5/C6): slice sorted arrays by partition
+            counts_d = ops.partition_counts(sk, self.world)
+            lens, blob = ops.extract_words(text, sp)
+            # per-partition blob byte counts: cumsum(lens) at boundaries
+            bnd = torch.cumsum(counts_d, 0)
+            if lens.numel():
+                lcs = torch.cumsum(lens, 0)
+                cum = torch.where(
+                    bnd > 0, lcs.index_select(0, (bnd - 1).clamp(min=0)),
+                    torch.zeros_like(bnd))
+                blob_counts_d = torch.cat([cum[:1], cum[1:] - cum[:-1]])
+            else:
+                blob_counts_d = torch.zeros_like(counts_d)
+            recv_counts_d = dx.exchange_counts(counts_d, self.group)
+            recv_blob_d = dx.exchange_counts(blob_counts_d, self.group)
+            send_c = counts_d.cpu().tolist()
+            recv_c = recv_counts_d.cpu().tolist()
+            send_b = blob_counts_d.cpu().tolist()
+            recv_b = recv_blob_d.cpu().tolist()
+            rk = dx.exchange(sk, send_c, recv_c, self.group)
+            rv = dx.exchange(sv, send_c, recv_c, self.group)
+            rlens = dx.exchange(lens, send_c, recv_c, self.group)
+            rblob = dx.exchange(blob, send_b, recv_b, self.group)
+            # rebuild packed positions into the received blob
+            roff = torch.cumsum(rlens, 0) - rlens
+            rpos = (roff << 16) | rlens
+
+            # ---- REDUCE: sort received runs, segment, first-exemplar
+            k2, v2, p2 = ops.sort_by_key(rk, rv, rpos)
+            fk, fv, fp, _ = ops.reduce_by_key_sorted(k2, v2, p2)
+            blob_src = rblob
+        else:
+            fk, fv, fp = sk, sv, sp
+            blob_src = text
+
+        nwords = int(self._nwords.item())
+        return WordCountResult(keys=fk, counts=fv, pos=fp,
+                               blob_src=blob_src, nwords=nwords)

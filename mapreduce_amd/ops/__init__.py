@@ -1,0 +1,228 @@
+"""High-level wrappers over the CDNA4 HIP kernels (mapreduce_amd._hip_ops).
+
+The HIP path is mandatory on GPU: if the extension is missing on a machine
+with a visible GPU we raise immediately instead of silently falling back to
+eager PyTorch (the fallbacks in this package exist only for CPU-only test
+environments).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import torch
+
+_ext = None
+_ext_err: Optional[BaseException] = None
+try:
+    from mapreduce_amd import _hip_ops as _ext  # built by setup.py, in-tree
+except Exception as e:  # pragma: no cover
+    _ext_err = e
+
+
+def have_ext() -> bool:
+    return _ext is not None
+
+
+def ext():
+    if _ext is None:
+        raise ImportError(
+            "mapreduce_amd._hip_ops is not built; run "
+            "`PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace`. "
+            f"(import error: {_ext_err})")
+    return _ext
+
+
+def require_gpu_ext() -> None:
+    """Loud failure when a GPU is present but the HIP extension is not —
+    GPU runs must never silently use an eager fallback."""
+    if torch.cuda.is_available():
+        ext()
+
+
+def _next_pow2(n: int) -> int:
+    return 1 << max(4, math.ceil(math.log2(max(n, 2))))
+
+
+# ---------------------------------------------------------------------------
+# tokenizer (K2/K3)
+# ---------------------------------------------------------------------------
+
+
+def tokenize_words(text: torch.Tensor):
+    """text: uint8 tensor -> (hashes i64[n], pos i64[n], n).
+
+    A word needs >=1 byte + separator, so capacity (len+1)//2 + 1 never
+    overflows.  Synchronizes once to read back the word count.
+    """
+    if not text.is_cuda:
+        from . import _cpu
+        return _cpu.tokenize_words(text)
+    cap = text.numel() // 2 + 16
+    h, p, c = ext().tokenize(text, cap)
+    n = int(c.item())
+    assert n <= cap
+    return h[:n], p[:n], n
+
+
+# ---------------------------------------------------------------------------
+# hash-table combiner (K5 aggregation form)
+# ---------------------------------------------------------------------------
+
+
+class HashTable:
+    """Open-addressing (u64 key -> i64 sum) table with optional exemplar
+    tracking.  Lives in HBM; sized as next_pow2(2 * expected_uniques)."""
+
+    def __init__(self, capacity: int, device, exemplar: bool = True):
+        cap = _next_pow2(2 * capacity)
+        opts = dict(device=device, dtype=torch.int64)
+        self.tkeys = torch.full((cap,), -1, **opts)  # -1 bits == HT_EMPTY
+        self.tvals = torch.zeros((cap,), **opts)
+        self.texm = torch.zeros((cap if exemplar else 0,), **opts)
+        self.cap = cap
+
+    def tokenize_count(self, text: torch.Tensor, pos_base: int,
+                       nwords: torch.Tensor):
+        """Fused map+combine: tokenize `text` (a split, offset pos_base in
+        the rank corpus) straight into the table; nwords (i64[1] on device)
+        accumulates the word count without a host sync."""
+        ext().tokenize_count(text, pos_base, self.tkeys, self.tvals,
+                             self.texm, nwords)
+
+    def insert_count(self, keys: torch.Tensor, pos: Optional[torch.Tensor]):
+        n = keys.numel()
+        if n == 0:
+            return
+        p = pos if pos is not None else torch.empty(0, dtype=torch.int64,
+                                                    device=keys.device)
+        ext().hash_insert_count(keys, p, self.tkeys, self.tvals, self.texm, n)
+
+    def insert_sum(self, keys: torch.Tensor, vals: torch.Tensor):
+        n = keys.numel()
+        if n == 0:
+            return
+        ext().hash_insert_sum_i64(keys, vals, self.tkeys, self.tvals, n)
+
+    def extract(self):
+        """-> (keys, vals, pos) compacted, unsorted.  One sync for count."""
+        k, v, p, c = ext().hash_extract(self.tkeys, self.tvals, self.texm)
+        n = int(c.item())
+        return k[:n], v[:n], (p[:n] if p.numel() else p)
+
+
+# ---------------------------------------------------------------------------
+# radix sort (K1)
+# ---------------------------------------------------------------------------
+
+
+def make_table(capacity: int, device, exemplar: bool = True):
+    """Hash-table combiner for the given device (GPU: HIP kernels; CPU:
+    test-tier dict)."""
+    if torch.device(device).type == "cuda":
+        return HashTable(capacity, device, exemplar)
+    from ._cpu import CpuHashTable
+    return CpuHashTable(capacity, device, exemplar)
+
+
+def sort_pairs(keys: torch.Tensor, vals: Optional[torch.Tensor] = None,
+               bits: int = 64):
+    """Stable LSD radix sort of u64 keys (bit-pattern order) w/ payload."""
+    if not keys.is_cuda:
+        from . import _cpu
+        return _cpu.sort_pairs(keys, vals, bits)
+    empty = torch.empty(0, dtype=torch.int64, device=keys.device)
+    k, v = ext().radix_sort_pairs(keys.contiguous(),
+                                  vals.contiguous() if vals is not None
+                                  else empty, bits)
+    return (k, v if vals is not None else None)
+
+
+def sort_by_key(keys: torch.Tensor, *others: torch.Tensor, bits: int = 64):
+    """Sort keys; reorder any number of same-length tensors alongside."""
+    if not others:
+        k, _ = sort_pairs(keys, None, bits)
+        return (k,)
+    idx = torch.arange(keys.numel(), device=keys.device, dtype=torch.int64)
+    k, perm = sort_pairs(keys, idx, bits)
+    return (k,) + tuple(t.index_select(0, perm) for t in others)
+
+
+# ---------------------------------------------------------------------------
+# segmented reduce-by-key over sorted keys (K4+K5 sorted form)
+# ---------------------------------------------------------------------------
+
+
+def reduce_by_key_sorted(keys: torch.Tensor,
+                         vals: Optional[torch.Tensor] = None,
+                         aux: Optional[torch.Tensor] = None):
+    """keys sorted (u64 bit order); vals i64/f64 or None (=count 1s);
+    aux: optional per-element u64 whose first value per segment is kept
+    (exemplar positions).  Returns (ukeys, reduced, uaux?, nseg)."""
+    if not keys.is_cuda:
+        from . import _cpu
+        return _cpu.reduce_by_key_sorted(keys, vals, aux)
+    n = keys.numel()
+    if n == 0:
+        z = torch.empty(0, dtype=torch.int64, device=keys.device)
+        return z, z, (z if aux is not None else None), 0
+    flags = ext().head_flags(keys)
+    seg = torch.cumsum(flags, 0)
+    nseg = int(seg[-1].item())
+    if vals is None or vals.dtype == torch.int64:
+        v = vals if vals is not None else torch.empty(
+            0, dtype=torch.int64, device=keys.device)
+        uk, uv = ext().seg_reduce_i64(keys, v, seg, nseg)
+    elif vals.dtype == torch.float64:
+        uk, uv = ext().seg_reduce_f64(keys, vals, seg, nseg)
+    else:
+        raise TypeError(f"unsupported value dtype {vals.dtype}")
+    ua = ext().seg_first_u64(aux, seg, nseg) if aux is not None else None
+    return uk, uv, ua, nseg
+
+
+def segment_boundaries(keys: torch.Tensor):
+    """-> (seg i64[n] 1-based segment ids, nseg). keys sorted."""
+    if not keys.is_cuda:
+        from . import _cpu
+        return _cpu.segment_boundaries(keys)
+    flags = ext().head_flags(keys)
+    seg = torch.cumsum(flags, 0)
+    nseg = int(seg[-1].item()) if keys.numel() else 0
+    return seg, nseg
+
+
+# ---------------------------------------------------------------------------
+# partitioning (K2) — send layout for the RCCL all-to-all
+# ---------------------------------------------------------------------------
+
+
+def partition_counts(keys: torch.Tensor, nparts: int) -> torch.Tensor:
+    """Histogram of partition_of(h) = mulhi(h, nparts).  When keys are
+    sorted, partitions are contiguous, so cumsum(hist) gives the slice
+    offsets for all-to-all send buffers."""
+    if not keys.is_cuda:
+        from . import _cpu
+        return _cpu.partition_counts(keys, nparts)
+    return ext().partition_hist(keys, nparts)
+
+
+# ---------------------------------------------------------------------------
+# dictionary extraction (K7/K8)
+# ---------------------------------------------------------------------------
+
+
+def extract_words(text: torch.Tensor, pos: torch.Tensor):
+    """-> (lens i64[n], blob u8[sum lens]): the exemplar word bytes packed
+    back-to-back, for the hash -> string dictionary."""
+    if not text.is_cuda:
+        from . import _cpu
+        return _cpu.extract_words(text, pos)
+    lens = ext().pos_len(pos)
+    if pos.numel() == 0:
+        return lens, torch.empty(0, dtype=torch.uint8, device=text.device)
+    offs = torch.cumsum(lens, 0) - lens
+    total = int((offs[-1] + lens[-1]).item())
+    blob = ext().gather_bytes(text, pos, offs, total)
+    return lens, blob

@@ -1,0 +1,146 @@
+"""CPU reference implementations of the HIP ops.
+
+Test tier only: lets the distributed engine run end-to-end on CPU (gloo,
+world_size > 1) in environments without a GPU, and doubles as the numerics
+oracle the HIP kernels are tested against.  On a machine with a GPU these
+are never silently used (ops.require_gpu_ext)."""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import numpy as np
+import torch
+
+from ..utils.tuple import FNV64_OFFSET, FNV64_PRIME
+
+_MASK64 = (1 << 64) - 1
+HT_EMPTY = _MASK64
+_WS = frozenset(b" \t\n\v\f\r")
+
+
+def _u64(t: torch.Tensor) -> np.ndarray:
+    return t.numpy().view(np.uint64)
+
+
+def _from_u64(a: np.ndarray, like: Optional[torch.Tensor] = None):
+    return torch.from_numpy(a.astype(np.uint64, copy=False).view(np.int64))
+
+
+def tokenize_words(text: torch.Tensor):
+    data = bytes(text.numpy().tobytes())
+    hashes = []
+    pos = []
+    n = len(data)
+    i = 0
+    while i < n:
+        if data[i] in _WS:
+            i += 1
+            continue
+        j = i
+        h = FNV64_OFFSET
+        while j < n and data[j] not in _WS:
+            h = ((h ^ data[j]) * FNV64_PRIME) & _MASK64
+            j += 1
+        hashes.append(h)
+        pos.append((i << 16) | min(j - i, 0xFFFF))
+        i = j
+    k = _from_u64(np.array(hashes, dtype=np.uint64))
+    p = _from_u64(np.array(pos, dtype=np.uint64))
+    return k, p, len(hashes)
+
+
+class CpuHashTable:
+    def __init__(self, capacity: int, device=None, exemplar: bool = True):
+        self._counts = {}
+        self._exm = {} if exemplar else None
+
+    def tokenize_count(self, text: torch.Tensor, pos_base: int,
+                       nwords: torch.Tensor):
+        k, p, n = tokenize_words(text)
+        if pos_base:
+            pu = _u64(p) + (np.uint64(pos_base) << np.uint64(16))
+            p = _from_u64(pu)
+        self.insert_count(k, p)
+        nwords += n
+
+    def insert_count(self, keys: torch.Tensor, pos: Optional[torch.Tensor]):
+        ku = _u64(keys)
+        pu = _u64(pos) if pos is not None and pos.numel() else None
+        for i, k in enumerate(ku.tolist()):
+            k = HT_EMPTY - 1 if k == HT_EMPTY else k
+            if k not in self._counts:
+                self._counts[k] = 0
+                if self._exm is not None and pu is not None:
+                    self._exm[k] = int(pu[i])
+            self._counts[k] += 1
+
+    def insert_sum(self, keys: torch.Tensor, vals: torch.Tensor):
+        for k, v in zip(_u64(keys).tolist(), vals.tolist()):
+            k = HT_EMPTY - 1 if k == HT_EMPTY else k
+            self._counts[k] = self._counts.get(k, 0) + v
+
+    def extract(self):
+        ks = np.array(list(self._counts.keys()), dtype=np.uint64)
+        vs = torch.tensor(list(self._counts.values()), dtype=torch.int64)
+        if self._exm is not None:
+            ps = np.array([self._exm.get(int(k), 0) for k in ks],
+                          dtype=np.uint64)
+            return _from_u64(ks), vs, _from_u64(ps)
+        return _from_u64(ks), vs, torch.empty(0, dtype=torch.int64)
+
+
+def sort_pairs(keys: torch.Tensor, vals: Optional[torch.Tensor], bits: int):
+    ku = _u64(keys)
+    order = np.argsort(ku, kind="stable")
+    sk = _from_u64(ku[order])
+    if vals is None or vals.numel() == 0:
+        return sk, None
+    return sk, vals[torch.from_numpy(order.astype(np.int64))]
+
+
+def reduce_by_key_sorted(keys, vals, aux):
+    ku = _u64(keys)
+    ukeys, idx = np.unique(ku, return_index=True)
+    if vals is None:
+        sums = np.add.reduceat(np.ones(len(ku), dtype=np.int64), idx)
+        uv = torch.from_numpy(sums)
+    else:
+        uv = torch.from_numpy(np.add.reduceat(vals.numpy(), idx))
+    ua = None
+    if aux is not None:
+        ua = _from_u64(_u64(aux)[idx])
+    return _from_u64(ukeys), uv, ua, len(ukeys)
+
+
+def segment_boundaries(keys):
+    ku = _u64(keys)
+    if len(ku) == 0:
+        return torch.empty(0, dtype=torch.int64), 0
+    flags = np.ones(len(ku), dtype=np.int64)
+    flags[1:] = ku[1:] != ku[:-1]
+    seg = np.cumsum(flags)
+    return torch.from_numpy(seg), int(seg[-1])
+
+
+def partition_counts(keys: torch.Tensor, nparts: int) -> torch.Tensor:
+    ku = _u64(keys)
+    out = np.zeros(nparts, dtype=np.int64)
+    for k in ku.tolist():
+        out[(k * nparts) >> 64] += 1
+    return torch.from_numpy(out)
+
+
+def extract_words(text: torch.Tensor, pos: torch.Tensor):
+    data = bytes(text.numpy().tobytes())
+    pu = _u64(pos).tolist()
+    lens = []
+    chunks = []
+    for p in pu:
+        start, ln = p >> 16, p & 0xFFFF
+        lens.append(ln)
+        chunks.append(data[start:start + ln])
+    blob = b"".join(chunks)
+    return (torch.tensor(lens, dtype=torch.int64),
+            torch.from_numpy(np.frombuffer(blob, dtype=np.uint8).copy())
+            if blob else torch.empty(0, dtype=torch.uint8))

@@ -1,0 +1,52 @@
+// Common device helpers for the MapReduce CDNA4 kernels (gfx950 only).
+// Wavefront = 64 lanes; block sizes are multiples of 64 throughout.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+#define WAVE 64
+#define DEV __device__ __forceinline__
+
+typedef uint64_t u64;
+typedef uint32_t u32;
+typedef uint8_t u8;
+typedef int64_t i64;
+
+// Sentinel for empty hash-table slots.  A real FNV-1a hash could in
+// principle equal this; inserts remap it (see hash_table.hip) so the table
+// stays correct for every input.
+#define HT_EMPTY 0xFFFFFFFFFFFFFFFFull
+
+DEV u64 mulhi_u64(u64 a, u64 b) { return __umul64hi(a, b); }
+
+// partition id of a 64-bit hash for P partitions: floor(h * P / 2^64).
+// Monotonic in h, so a hash-sorted array is partition-contiguous — the
+// property that lets the RCCL all-to-all send buffers be plain slices.
+DEV u32 partition_of(u64 h, u32 nparts) { return (u32)mulhi_u64(h, (u64)nparts); }
+
+// FNV-1a 64 (must match mapreduce_amd.utils.tuple.fnv1a64)
+#define FNV64_OFFSET 0xCBF29CE484222325ull
+#define FNV64_PRIME 0x100000001B3ull
+
+DEV bool is_ws(u8 c) {
+  // Python str.split() whitespace set: \t \n \v \f \r ' '
+  return c == ' ' || (c >= 9 && c <= 13);
+}
+
+DEV u64 lane_id() { return __lane_id(); }
+
+// inclusive wave sum over 64 lanes
+DEV i64 wave_sum_i64(i64 v) {
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, WAVE);
+  return v;  // valid in lane 0
+}
+
+#define HIP_CHECK(expr)                                                     \
+  do {                                                                      \
+    hipError_t _e = (expr);                                                 \
+    if (_e != hipSuccess) {                                                 \
+      printf("HIP error %s at %s:%d\n", hipGetErrorString(_e), __FILE__,    \
+             __LINE__);                                                     \
+    }                                                                       \
+  } while (0)

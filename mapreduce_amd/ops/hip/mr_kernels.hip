@@ -1,0 +1,324 @@
+// MapReduce hot-path kernels for MI355X (gfx950), CDNA4-native.
+//
+// These replace the reference's CPU-Lua hot loops:
+//   K2/K3  tokenize_kernel        — word boundary scan + FNV-1a64 key hash
+//   K5     hash_insert/extract    — hash-table combiner / reduce-by-key
+//          (aggregation form; the sort form is radix_sort.hip + segmented
+//          reduce below)
+//   K5     seg_reduce kernels     — segmented reduce-by-key over sorted runs
+//   K2     partition_hist         — all-to-all send counts (C5 setup)
+//   K7/K8  gather_bytes           — exemplar word extraction for the
+//          hash -> string dictionary at the finalfn boundary
+//
+// All memory-bound: vectorized accesses, grid-stride loops, one atomic per
+// wave where aggregation applies (guide G12/G13).
+
+#include "common.h"
+
+// ---------------------------------------------------------------------------
+// K2/K3: tokenizer — emit (hash64, pos) per word of a byte buffer
+// ---------------------------------------------------------------------------
+// pos packs (start << 16 | len) so the exemplar word bytes can be gathered
+// later; output order is nondeterministic (atomic append) — every consumer
+// sorts or hash-aggregates, so order never matters.
+
+#define TOK_BYTES 16
+
+__global__ void tokenize_kernel(const u8* __restrict__ text, long n,
+                                u64* __restrict__ out_hash,
+                                u64* __restrict__ out_pos,
+                                unsigned long long* __restrict__ counter,
+                                long cap) {
+  long t0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * (long)TOK_BYTES;
+  long stride = (long)gridDim.x * blockDim.x * TOK_BYTES;
+  for (long base = t0; base < n; base += stride) {
+    long end = base + TOK_BYTES < n ? base + TOK_BYTES : n;
+    u8 prev = (base == 0) ? ' ' : text[base - 1];
+    for (long i = base; i < end; ++i) {
+      u8 c = text[i];
+      if (!is_ws(c) && is_ws(prev)) {
+        u64 h = FNV64_OFFSET;
+        long j = i;
+        while (j < n) {
+          u8 cc = text[j];
+          if (is_ws(cc)) break;
+          h ^= cc;
+          h *= FNV64_PRIME;
+          ++j;
+        }
+        unsigned long long idx = atomicAdd(counter, 1ull);
+        if ((long)idx < cap) {
+          out_hash[idx] = h;
+          long len = j - i;
+          if (len > 0xFFFF) len = 0xFFFF;
+          out_pos[idx] = ((u64)i << 16) | (u64)len;
+        }
+      }
+      prev = c;
+    }
+  }
+}
+
+// hash-table primitives (shared by the fused and unfused insert paths)
+DEV u64 remap_key(u64 k) {
+  // keep HT_EMPTY available as the slot sentinel for arbitrary inputs
+  return k == HT_EMPTY ? HT_EMPTY - 1 : k;
+}
+
+DEV u32 first_slot(u64 k, u64 cap_mask) {
+  // keys are already FNV-mixed; fold high bits so table size bits differ
+  return (u32)((k ^ (k >> 32)) & cap_mask);
+}
+
+// ---------------------------------------------------------------------------
+// K2+K5 fused: tokenize-and-count straight into the hash table.
+// One pass over the text, no intermediate (hash, pos) arrays, no host sync
+// per split (the word counter is read once per phase) — the map emit and
+// the inline combiner of job.lua:83-97 collapsed into a single kernel.
+// pos_base offsets word positions so many splits share one corpus buffer.
+// ---------------------------------------------------------------------------
+
+__device__ void ht_count_one(u64 k, u64 p, u64* tkeys, i64* tvals, u64* texm,
+                             u64 cap_mask);
+
+__global__ void tokenize_count_kernel(const u8* __restrict__ text, long n,
+                                      u64 pos_base,
+                                      u64* __restrict__ tkeys,
+                                      i64* __restrict__ tvals,
+                                      u64* __restrict__ texm, u64 cap_mask,
+                                      unsigned long long* __restrict__ nwords) {
+  long t0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * (long)TOK_BYTES;
+  long stride = (long)gridDim.x * blockDim.x * TOK_BYTES;
+  unsigned long long my_words = 0;
+  for (long base = t0; base < n; base += stride) {
+    long end = base + TOK_BYTES < n ? base + TOK_BYTES : n;
+    u8 prev = (base == 0) ? ' ' : text[base - 1];
+    for (long i = base; i < end; ++i) {
+      u8 c = text[i];
+      if (!is_ws(c) && is_ws(prev)) {
+        u64 h = FNV64_OFFSET;
+        long j = i;
+        while (j < n) {
+          u8 cc = text[j];
+          if (is_ws(cc)) break;
+          h ^= cc;
+          h *= FNV64_PRIME;
+          ++j;
+        }
+        long len = j - i;
+        if (len > 0xFFFF) len = 0xFFFF;
+        u64 p = ((pos_base + (u64)i) << 16) | (u64)len;
+        ht_count_one(remap_key(h), p, tkeys, tvals, texm, cap_mask);
+        ++my_words;
+      }
+      prev = c;
+    }
+  }
+  // one atomic per wave for the word-count metric (guide G12)
+  unsigned long long ws = my_words;
+  for (int off = 32; off > 0; off >>= 1) ws += __shfl_down(ws, off, WAVE);
+  if (lane_id() == 0 && ws) atomicAdd(nwords, ws);
+}
+
+// ---------------------------------------------------------------------------
+// K5 (aggregation form): open-addressing hash table, linear probing.
+// Combiner for declared associative+commutative reducers (job.lua:104-106 —
+// the reference's own fast-path flags select this path).
+// ---------------------------------------------------------------------------
+// Table: tkeys (HT_EMPTY = free), tvals (i64 sum), texm (first-inserter
+// exemplar pos; 0 if untracked).  cap is a power of two.
+
+__device__ void ht_count_one(u64 k, u64 p, u64* tkeys, i64* tvals, u64* texm,
+                             u64 cap_mask) {
+  u32 slot = first_slot(k, cap_mask);
+  while (true) {
+    u64 cur = tkeys[slot];
+    if (cur == k) {
+      atomicAdd((unsigned long long*)&tvals[slot], 1ull);
+      return;
+    }
+    if (cur == HT_EMPTY) {
+      u64 prev = atomicCAS((unsigned long long*)&tkeys[slot],
+                           (unsigned long long)HT_EMPTY,
+                           (unsigned long long)k);
+      if (prev == HT_EMPTY) {
+        if (texm) texm[slot] = p;  // first inserter records the exemplar;
+                                   // consumed only after kernel completion
+        atomicAdd((unsigned long long*)&tvals[slot], 1ull);
+        return;
+      }
+      if (prev == k) {
+        atomicAdd((unsigned long long*)&tvals[slot], 1ull);
+        return;
+      }
+    }
+    slot = (u32)((slot + 1) & cap_mask);
+  }
+}
+
+__global__ void hash_insert_count_kernel(const u64* __restrict__ keys,
+                                         const u64* __restrict__ pos,
+                                         long n, u64* __restrict__ tkeys,
+                                         i64* __restrict__ tvals,
+                                         u64* __restrict__ texm,
+                                         u64 cap_mask) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride)
+    ht_count_one(remap_key(keys[i]), pos ? pos[i] : 0, tkeys, tvals, texm,
+                 cap_mask);
+}
+
+// generic (key, i64 value) insert — gradient counts, inverted-index sizes...
+__global__ void hash_insert_sum_i64_kernel(const u64* __restrict__ keys,
+                                           const i64* __restrict__ vals,
+                                           long n, u64* __restrict__ tkeys,
+                                           i64* __restrict__ tvals,
+                                           u64 cap_mask) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    u64 k = remap_key(keys[i]);
+    i64 v = vals[i];
+    u32 slot = first_slot(k, cap_mask);
+    while (true) {
+      u64 cur = tkeys[slot];
+      if (cur == k) {
+        atomicAdd((unsigned long long*)&tvals[slot], (unsigned long long)v);
+        break;
+      }
+      if (cur == HT_EMPTY) {
+        u64 prev = atomicCAS((unsigned long long*)&tkeys[slot],
+                             (unsigned long long)HT_EMPTY,
+                             (unsigned long long)k);
+        if (prev == HT_EMPTY || prev == k) {
+          atomicAdd((unsigned long long*)&tvals[slot], (unsigned long long)v);
+          break;
+        }
+      }
+      slot = (u32)((slot + 1) & cap_mask);
+    }
+  }
+}
+
+__global__ void hash_extract_kernel(const u64* __restrict__ tkeys,
+                                    const i64* __restrict__ tvals,
+                                    const u64* __restrict__ texm, long cap,
+                                    u64* __restrict__ okeys,
+                                    i64* __restrict__ ovals,
+                                    u64* __restrict__ opos,
+                                    unsigned long long* __restrict__ counter) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < cap; i += stride) {
+    u64 k = tkeys[i];
+    if (k != HT_EMPTY) {
+      unsigned long long idx = atomicAdd(counter, 1ull);
+      okeys[idx] = k;
+      ovals[idx] = tvals[i];
+      if (opos && texm) opos[idx] = texm[i];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// K5 (sorted form): segmented reduce-by-key over a key-sorted run.
+// head_flags + (host cumsum) + scatter stage; K4's k-way merge becomes
+// "sort once, then segment" (SURVEY.md K4).
+// ---------------------------------------------------------------------------
+
+__global__ void head_flags_kernel(const u64* __restrict__ keys, long n,
+                                  i64* __restrict__ flags) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride)
+    flags[i] = (i == 0) || (keys[i] != keys[i - 1]);
+}
+
+// seg[i] = (inclusive cumsum of flags)[i] - 1; scatter keys at heads, add vals
+__global__ void seg_scatter_i64_kernel(const u64* __restrict__ keys,
+                                       const i64* __restrict__ vals,
+                                       const i64* __restrict__ seg, long n,
+                                       u64* __restrict__ okeys,
+                                       i64* __restrict__ ovals) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    i64 s = seg[i] - 1;
+    if (i == 0 || seg[i] != seg[i - 1]) okeys[s] = keys[i];
+    atomicAdd((unsigned long long*)&ovals[s],
+              (unsigned long long)(vals ? vals[i] : 1));
+  }
+}
+
+// pick one auxiliary value (e.g. exemplar pos) per segment: first element
+__global__ void seg_first_u64_kernel(const u64* __restrict__ aux,
+                                     const i64* __restrict__ seg, long n,
+                                     u64* __restrict__ oaux) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride)
+    if (i == 0 || seg[i] != seg[i - 1]) oaux[seg[i] - 1] = aux[i];
+}
+
+__global__ void seg_scatter_f64_kernel(const u64* __restrict__ keys,
+                                       const double* __restrict__ vals,
+                                       const i64* __restrict__ seg, long n,
+                                       u64* __restrict__ okeys,
+                                       double* __restrict__ ovals) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    i64 s = seg[i] - 1;
+    if (i == 0 || seg[i] != seg[i - 1]) okeys[s] = keys[i];
+    atomicAdd(&ovals[s], vals[i]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// K2: partition histogram — send counts for the RCCL all-to-all (C5)
+// ---------------------------------------------------------------------------
+
+#define MAX_PARTS 1024
+
+__global__ void partition_hist_kernel(const u64* __restrict__ keys, long n,
+                                      u32 nparts, i64* __restrict__ hist) {
+  __shared__ i64 lh[MAX_PARTS];
+  for (u32 p = threadIdx.x; p < nparts; p += blockDim.x) lh[p] = 0;
+  __syncthreads();
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride)
+    atomicAdd((unsigned long long*)&lh[partition_of(keys[i], nparts)], 1ull);
+  __syncthreads();
+  for (u32 p = threadIdx.x; p < nparts; p += blockDim.x)
+    if (lh[p]) atomicAdd((unsigned long long*)&hist[p],
+                         (unsigned long long)lh[p]);
+}
+
+// ---------------------------------------------------------------------------
+// K7/K8: gather exemplar word bytes into a packed blob (dictionary build)
+// ---------------------------------------------------------------------------
+
+__global__ void gather_bytes_kernel(const u8* __restrict__ text,
+                                    const u64* __restrict__ pos,
+                                    const i64* __restrict__ out_off, long n,
+                                    u8* __restrict__ out) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    u64 p = pos[i];
+    long start = (long)(p >> 16);
+    long len = (long)(p & 0xFFFF);
+    long o = out_off[i];
+    for (long j = 0; j < len; ++j) out[o + j] = text[start + j];
+  }
+}
+
+// lengths from packed pos
+__global__ void pos_len_kernel(const u64* __restrict__ pos, long n,
+                               i64* __restrict__ lens) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) lens[i] = (i64)(pos[i] & 0xFFFF);
+}

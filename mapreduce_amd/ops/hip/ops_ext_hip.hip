@@ -1,0 +1,259 @@
+#include "hip/hip_runtime.h"
+// Torch extension glue for the MapReduce CDNA4 kernels (single TU).
+//
+// Tensors use int64 storage for 64-bit keys; kernels reinterpret the bits as
+// u64 (all ordering/partitioning comparisons happen inside kernels, so
+// torch's signed view never matters).  Everything launches on the current
+// HIP stream; nothing here synchronizes except the explicitly named *_count
+// readbacks, which callers do lazily.
+
+#include <ATen/hip/HIPContext.h>
+#include <torch/extension.h>
+
+#include "mr_kernels.hip"
+#include "radix_sort.hip"
+
+namespace {
+
+constexpr int kBlock = 256;
+// G11: memory-bound grid-stride kernels cap the grid and stride the rest
+constexpr long kMaxBlocks = 2048;
+
+long grid_for(long n, long per_thread = 1) {
+  long want = (n + (long)kBlock * per_thread - 1) / ((long)kBlock * per_thread);
+  if (want < 1) want = 1;
+  return want < kMaxBlocks ? want : kMaxBlocks;
+}
+
+hipStream_t cur_stream() { return at::hip::getCurrentHIPStream(); }
+
+u64* u64p(torch::Tensor& t) { return reinterpret_cast<u64*>(t.data_ptr<i64>()); }
+const u64* u64cp(const torch::Tensor& t) {
+  return reinterpret_cast<const u64*>(t.data_ptr<i64>());
+}
+
+void check_dev_i64(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kInt64, name, " must be int64");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------- K2/K3
+std::vector<torch::Tensor> tokenize(torch::Tensor text, long cap) {
+  TORCH_CHECK(text.is_cuda() && text.scalar_type() == torch::kUInt8 &&
+              text.is_contiguous(), "text must be contiguous u8 on GPU");
+  long n = text.numel();
+  auto opts = torch::TensorOptions().device(text.device()).dtype(torch::kInt64);
+  auto out_hash = torch::empty({cap}, opts);
+  auto out_pos = torch::empty({cap}, opts);
+  auto counter = torch::zeros({1}, opts);
+  long blocks = grid_for(n, TOK_BYTES);
+  hipLaunchKernelGGL(tokenize_kernel, dim3(blocks), dim3(kBlock), 0,
+                     cur_stream(), text.data_ptr<u8>(), n, u64p(out_hash),
+                     u64p(out_pos),
+                     reinterpret_cast<unsigned long long*>(counter.data_ptr<i64>()),
+                     cap);
+  return {out_hash, out_pos, counter};
+}
+
+// ------------------------------------------------------------------ K2+K5
+void tokenize_count(torch::Tensor text, long pos_base, torch::Tensor tkeys,
+                    torch::Tensor tvals, torch::Tensor texm,
+                    torch::Tensor nwords) {
+  TORCH_CHECK(text.is_cuda() && text.scalar_type() == torch::kUInt8 &&
+              text.is_contiguous(), "text must be contiguous u8 on GPU");
+  long n = text.numel();
+  long cap = tkeys.numel();
+  TORCH_CHECK((cap & (cap - 1)) == 0, "table capacity must be a power of 2");
+  if (!n) return;
+  hipLaunchKernelGGL(tokenize_count_kernel, dim3(grid_for(n, TOK_BYTES)),
+                     dim3(kBlock), 0, cur_stream(), text.data_ptr<u8>(), n,
+                     (u64)pos_base, u64p(tkeys), tvals.data_ptr<i64>(),
+                     texm.numel() ? u64p(texm) : nullptr, (u64)(cap - 1),
+                     reinterpret_cast<unsigned long long*>(nwords.data_ptr<i64>()));
+}
+
+// ----------------------------------------------------------------------- K5a
+void hash_insert_count(torch::Tensor keys, torch::Tensor pos,
+                       torch::Tensor tkeys, torch::Tensor tvals,
+                       torch::Tensor texm, long n) {
+  check_dev_i64(keys, "keys");
+  long cap = tkeys.numel();
+  TORCH_CHECK((cap & (cap - 1)) == 0, "table capacity must be a power of 2");
+  hipLaunchKernelGGL(hash_insert_count_kernel, dim3(grid_for(n)), dim3(kBlock),
+                     0, cur_stream(), u64cp(keys),
+                     pos.numel() ? u64cp(pos) : nullptr, n, u64p(tkeys),
+                     tvals.data_ptr<i64>(),
+                     texm.numel() ? u64p(texm) : nullptr, (u64)(cap - 1));
+}
+
+void hash_insert_sum_i64(torch::Tensor keys, torch::Tensor vals,
+                         torch::Tensor tkeys, torch::Tensor tvals, long n) {
+  check_dev_i64(keys, "keys");
+  long cap = tkeys.numel();
+  TORCH_CHECK((cap & (cap - 1)) == 0, "table capacity must be a power of 2");
+  hipLaunchKernelGGL(hash_insert_sum_i64_kernel, dim3(grid_for(n)),
+                     dim3(kBlock), 0, cur_stream(), u64cp(keys),
+                     vals.data_ptr<i64>(), n, u64p(tkeys),
+                     tvals.data_ptr<i64>(), (u64)(cap - 1));
+}
+
+std::vector<torch::Tensor> hash_extract(torch::Tensor tkeys,
+                                        torch::Tensor tvals,
+                                        torch::Tensor texm) {
+  long cap = tkeys.numel();
+  auto opts = tkeys.options();
+  auto okeys = torch::empty({cap}, opts);
+  auto ovals = torch::empty({cap}, opts);
+  bool exm = texm.numel() > 0;
+  auto opos = torch::empty({exm ? cap : 0}, opts);
+  auto counter = torch::zeros({1}, opts);
+  hipLaunchKernelGGL(hash_extract_kernel, dim3(grid_for(cap)), dim3(kBlock), 0,
+                     cur_stream(), u64cp(tkeys), tvals.data_ptr<i64>(),
+                     exm ? u64cp(texm) : nullptr, cap, u64p(okeys),
+                     ovals.data_ptr<i64>(), exm ? u64p(opos) : nullptr,
+                     reinterpret_cast<unsigned long long*>(counter.data_ptr<i64>()));
+  return {okeys, ovals, opos, counter};
+}
+
+// ----------------------------------------------------------------------- K5b
+torch::Tensor head_flags(torch::Tensor keys) {
+  check_dev_i64(keys, "keys");
+  long n = keys.numel();
+  auto flags = torch::empty({n}, keys.options());
+  if (n)
+    hipLaunchKernelGGL(head_flags_kernel, dim3(grid_for(n)), dim3(kBlock), 0,
+                       cur_stream(), u64cp(keys), n, flags.data_ptr<i64>());
+  return flags;
+}
+
+std::vector<torch::Tensor> seg_reduce_i64(torch::Tensor keys,
+                                          torch::Tensor vals,
+                                          torch::Tensor seg, long nseg) {
+  check_dev_i64(keys, "keys");
+  long n = keys.numel();
+  auto okeys = torch::empty({nseg}, keys.options());
+  auto ovals = torch::zeros({nseg}, keys.options());
+  if (n)
+    hipLaunchKernelGGL(seg_scatter_i64_kernel, dim3(grid_for(n)), dim3(kBlock),
+                       0, cur_stream(), u64cp(keys),
+                       vals.numel() ? vals.data_ptr<i64>() : nullptr,
+                       seg.data_ptr<i64>(), n, u64p(okeys),
+                       ovals.data_ptr<i64>());
+  return {okeys, ovals};
+}
+
+torch::Tensor seg_first_u64(torch::Tensor aux, torch::Tensor seg, long nseg) {
+  long n = aux.numel();
+  auto oaux = torch::zeros({nseg}, aux.options());
+  if (n)
+    hipLaunchKernelGGL(seg_first_u64_kernel, dim3(grid_for(n)), dim3(kBlock),
+                       0, cur_stream(), u64cp(aux), seg.data_ptr<i64>(), n,
+                       u64p(oaux));
+  return oaux;
+}
+
+std::vector<torch::Tensor> seg_reduce_f64(torch::Tensor keys,
+                                          torch::Tensor vals,
+                                          torch::Tensor seg, long nseg) {
+  check_dev_i64(keys, "keys");
+  long n = keys.numel();
+  auto okeys = torch::empty({nseg}, keys.options());
+  auto ovals = torch::zeros({nseg}, vals.options());
+  if (n)
+    hipLaunchKernelGGL(seg_scatter_f64_kernel, dim3(grid_for(n)), dim3(kBlock),
+                       0, cur_stream(), u64cp(keys), vals.data_ptr<double>(),
+                       seg.data_ptr<i64>(), n, u64p(okeys),
+                       ovals.data_ptr<double>());
+  return {okeys, ovals};
+}
+
+// ------------------------------------------------------------------------ K2
+torch::Tensor partition_hist(torch::Tensor keys, long nparts) {
+  check_dev_i64(keys, "keys");
+  TORCH_CHECK(nparts <= MAX_PARTS, "nparts too large");
+  long n = keys.numel();
+  auto hist = torch::zeros({nparts}, keys.options());
+  if (n)
+    hipLaunchKernelGGL(partition_hist_kernel, dim3(grid_for(n)), dim3(kBlock),
+                       0, cur_stream(), u64cp(keys), n, (u32)nparts,
+                       hist.data_ptr<i64>());
+  return hist;
+}
+
+// --------------------------------------------------------------------- K7/K8
+torch::Tensor pos_len(torch::Tensor pos) {
+  long n = pos.numel();
+  auto lens = torch::empty({n}, pos.options());
+  if (n)
+    hipLaunchKernelGGL(pos_len_kernel, dim3(grid_for(n)), dim3(kBlock), 0,
+                       cur_stream(), u64cp(pos), n, lens.data_ptr<i64>());
+  return lens;
+}
+
+torch::Tensor gather_bytes(torch::Tensor text, torch::Tensor pos,
+                           torch::Tensor out_off, long total) {
+  long n = pos.numel();
+  auto out = torch::zeros({total},
+                          torch::TensorOptions().device(text.device())
+                              .dtype(torch::kUInt8));
+  if (n)
+    hipLaunchKernelGGL(gather_bytes_kernel, dim3(grid_for(n)), dim3(kBlock), 0,
+                       cur_stream(), text.data_ptr<u8>(), u64cp(pos),
+                       out_off.data_ptr<i64>(), n, out.data_ptr<u8>());
+  return out;
+}
+
+// ------------------------------------------------------------------------ K1
+std::vector<torch::Tensor> radix_sort_pairs(torch::Tensor keys,
+                                            torch::Tensor vals, int bits) {
+  check_dev_i64(keys, "keys");
+  long n = keys.numel();
+  bool has_vals = vals.numel() > 0;
+  if (n == 0) return {keys, vals};
+  TORCH_CHECK(bits >= 1 && bits <= 64, "bits in [1,64]");
+  int passes = (bits + 7) / 8;
+  long ntiles = (n + RS_TILE - 1) / RS_TILE;
+  auto opts = keys.options();
+  auto kbuf = torch::empty({n}, opts);
+  auto vbuf = has_vals ? torch::empty({n}, opts) : torch::empty({0}, opts);
+  auto hist = torch::empty({(long)RS_BINS * ntiles}, opts);
+
+  torch::Tensor kin = keys, vin = vals, kout = kbuf, vout = vbuf;
+  for (int p = 0; p < passes; ++p) {
+    int shift = p * 8;
+    hipLaunchKernelGGL(radix_hist_kernel, dim3(ntiles), dim3(RS_BLOCK), 0,
+                       cur_stream(), u64cp(kin), n, shift, ntiles,
+                       hist.data_ptr<i64>());
+    // exclusive scan over the digit-major flat histogram = base[d][t]
+    auto scanned = torch::cumsum(hist, 0);
+    auto base = scanned - hist;
+    hipLaunchKernelGGL(radix_scatter_kernel, dim3(ntiles), dim3(RS_BLOCK), 0,
+                       cur_stream(), u64cp(kin),
+                       has_vals ? u64cp(vin) : nullptr, n, shift, ntiles,
+                       base.data_ptr<i64>(), u64p(kout),
+                       has_vals ? u64p(vout) : nullptr);
+    std::swap(kin, kout);
+    if (has_vals) std::swap(vin, vout);
+  }
+  return {kin, vin};
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("tokenize", &tokenize, "tokenize text -> (hash, pos, count)");
+  m.def("tokenize_count", &tokenize_count,
+        "fused tokenize + hash-table count");
+  m.def("hash_insert_count", &hash_insert_count);
+  m.def("hash_insert_sum_i64", &hash_insert_sum_i64);
+  m.def("hash_extract", &hash_extract);
+  m.def("head_flags", &head_flags);
+  m.def("seg_reduce_i64", &seg_reduce_i64);
+  m.def("seg_first_u64", &seg_first_u64);
+  m.def("seg_reduce_f64", &seg_reduce_f64);
+  m.def("partition_hist", &partition_hist);
+  m.def("pos_len", &pos_len);
+  m.def("gather_bytes", &gather_bytes);
+  m.def("radix_sort_pairs", &radix_sort_pairs);
+}

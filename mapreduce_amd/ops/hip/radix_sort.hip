@@ -1,0 +1,105 @@
+// LSD radix sort for u64 keys (+ u64 payload), CDNA4-native.
+//
+// 8-bit digits, LDS digit histograms, stable tile-local ranking via
+// wave64 ballots (8 single-bit ballots build the same-digit lane mask; the
+// lowest lane of each digit group publishes the group count to LDS, a
+// 256-thread prefix orders waves deterministically).  Per pass:
+//   1. radix_hist_kernel     — per-tile 256-bin LDS histogram, written
+//                              digit-major: hist[d * ntiles + t]
+//   2. (host) exclusive scan of the flat [256 * ntiles] array — digit-major
+//      order makes the scan produce exactly base[d][t]
+//   3. radix_scatter_kernel  — stable scatter to base[d][t] + local rank
+//
+// Replaces the reference's table.sort + heap merge (job.lua:194,
+// utils.lua:206-271): sort once, then segment (K4 -> K1+K5).
+
+#include "common.h"
+
+#define RS_BLOCK 256
+#define RS_WAVES (RS_BLOCK / WAVE)
+#define RS_ITEMS 8
+#define RS_TILE (RS_BLOCK * RS_ITEMS)
+#define RS_BINS 256
+
+__global__ __launch_bounds__(RS_BLOCK) void radix_hist_kernel(
+    const u64* __restrict__ keys, long n, int shift, long ntiles,
+    i64* __restrict__ hist) {
+  __shared__ u32 lh[RS_BINS];
+  for (int b = threadIdx.x; b < RS_BINS; b += blockDim.x) lh[b] = 0;
+  __syncthreads();
+  long tile = blockIdx.x;
+  long base = tile * RS_TILE;
+  for (int r = 0; r < RS_ITEMS; ++r) {
+    long i = base + r * RS_BLOCK + threadIdx.x;
+    if (i < n) {
+      u32 d = (u32)((keys[i] >> shift) & 0xFF);
+      atomicAdd(&lh[d], 1u);
+    }
+  }
+  __syncthreads();
+  for (int b = threadIdx.x; b < RS_BINS; b += blockDim.x)
+    hist[(long)b * ntiles + tile] = (i64)lh[b];
+}
+
+__global__ __launch_bounds__(RS_BLOCK) void radix_scatter_kernel(
+    const u64* __restrict__ keys, const u64* __restrict__ vals, long n,
+    int shift, long ntiles, const i64* __restrict__ base_dx,
+    u64* __restrict__ okeys, u64* __restrict__ ovals) {
+  // wavecnt[w][d]: this round's digit-d count of wave w, then (after the
+  // prefix phase) the exclusive tile-rank base for wave w's digit-d items.
+  __shared__ u32 wavecnt[RS_WAVES][RS_BINS];
+  __shared__ u32 cnt_base[RS_BINS];  // digit counts of earlier rounds
+  for (int b = threadIdx.x; b < RS_BINS; b += blockDim.x) cnt_base[b] = 0;
+  long tile = blockIdx.x;
+  long tbase = tile * RS_TILE;
+  int wave = threadIdx.x / WAVE;
+  int lane = threadIdx.x % WAVE;
+  u64 lt_mask = ((u64)1 << lane) - 1;
+
+  for (int r = 0; r < RS_ITEMS; ++r) {
+    for (int b = threadIdx.x; b < RS_BINS; b += blockDim.x)
+      for (int w = 0; w < RS_WAVES; ++w) wavecnt[w][b] = 0;
+    __syncthreads();
+
+    long i = tbase + (long)r * RS_BLOCK + threadIdx.x;
+    bool valid = i < n;
+    u64 k = valid ? keys[i] : 0;
+    u32 d = valid ? (u32)((k >> shift) & 0xFF) : 0;
+
+    // same-digit lane mask among valid lanes (8 single-bit ballots)
+    u64 m = __ballot(valid);
+    #pragma unroll
+    for (int b = 0; b < 8; ++b) {
+      u64 bb = __ballot(valid && ((d >> b) & 1));
+      m &= ((d >> b) & 1) ? bb : ~bb;
+    }
+    u32 lane_rank = 0;
+    if (valid) {
+      lane_rank = (u32)__popcll(m & lt_mask);
+      int leader = __ffsll((unsigned long long)m) - 1;
+      if (lane == leader) wavecnt[wave][d] = (u32)__popcll(m);
+    }
+    __syncthreads();
+
+    // deterministic wave order: thread b prefixes digit b over waves
+    if (threadIdx.x < RS_BINS) {
+      u32 run = cnt_base[threadIdx.x];
+      #pragma unroll
+      for (int w = 0; w < RS_WAVES; ++w) {
+        u32 c = wavecnt[w][threadIdx.x];
+        wavecnt[w][threadIdx.x] = run;
+        run += c;
+      }
+      cnt_base[threadIdx.x] = run;
+    }
+    __syncthreads();
+
+    if (valid) {
+      u32 tile_rank = wavecnt[wave][d] + lane_rank;
+      long pos = base_dx[(long)d * ntiles + tile] + tile_rank;
+      okeys[pos] = k;
+      if (ovals) ovals[pos] = vals[i];
+    }
+    __syncthreads();
+  }
+}

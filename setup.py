@@ -1,0 +1,33 @@
+"""Build the CDNA4 HIP extension in-tree (gfx950 only).
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+produces mapreduce_amd/_hip_ops*.so, which travels with the repo snapshot
+to GPU boxes (no JIT cache involved).
+"""
+
+import os
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from setuptools import setup
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+
+setup(
+    name="mapreduce_amd",
+    version="0.1.0",
+    packages=["mapreduce_amd"],
+    ext_modules=[
+        CUDAExtension(
+            name="mapreduce_amd._hip_ops",
+            sources=["mapreduce_amd/ops/hip/ops_ext.hip"],
+            extra_compile_args={
+                "cxx": ["-O3"],
+                "nvcc": ["-O3", "--offload-arch=gfx950", "-std=c++17"],
+            },
+        )
+    ],
+    cmdclass={"build_ext": BuildExtension},
+)

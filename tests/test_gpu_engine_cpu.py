@@ -1,0 +1,83 @@
+"""Engine-path tests on CPU: the exact code the GPU runs, exercised through
+the ops CPU fallbacks — single rank and gloo world_size=2 (the
+multi-process distributed shuffle, correct-by-construction before it ever
+touches RCCL)."""
+
+import collections
+import os
+import socket
+
+import pytest
+import torch
+
+from mapreduce_amd.gpu.corpus import make_corpus
+from mapreduce_amd.gpu.wordcount import WordCountJob
+
+
+def counter_oracle(text_bytes: bytes):
+    return collections.Counter(text_bytes.split())
+
+
+def test_corpus_shape():
+    c = make_corpus("cpu", nwords=10_000, nsplits=7, vocab_size=500, seed=3)
+    data = bytes(c.text.numpy().tobytes())
+    assert len(data.split()) == 10_000
+    assert len(c.split_offsets) == 8
+    # split boundaries are word boundaries (preceded by whitespace)
+    for off in c.split_offsets[1:-1]:
+        assert data[off - 1:off] == b" "
+
+
+def test_wordcount_job_single_rank_cpu():
+    c = make_corpus("cpu", nwords=20_000, nsplits=5, vocab_size=800, seed=1)
+    job = WordCountJob("cpu", vocab_estimate=2000)
+    res = job.run(c.text, c.splits())
+    assert res.nwords == 20_000
+    got = dict(res.to_host())
+    exp = counter_oracle(bytes(c.text.numpy().tobytes()))
+    assert got == dict(exp)
+
+
+def _dist_worker(rank, world, port, tmpdir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.distributed.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        c = make_corpus("cpu", nwords=8_000, nsplits=4, vocab_size=400,
+                        seed=100 + rank)
+        job = WordCountJob("cpu", vocab_estimate=1000)
+        res = job.run(c.text, c.splits())
+        pairs = res.to_host()
+        # each key must belong to this rank's partition (mulhi)
+        import numpy as np
+        keys = res.keys.numpy().view(np.uint64)
+        parts = ((keys.astype(object) * world) >> 64).astype(int)
+        assert (parts == rank).all()
+        # gather all (word, count) pairs to rank 0 and diff vs oracle
+        all_pairs = [None] * world
+        torch.distributed.all_gather_object(all_pairs, pairs)
+        all_texts = [None] * world
+        torch.distributed.all_gather_object(
+            all_texts, bytes(c.text.numpy().tobytes()))
+        if rank == 0:
+            got = collections.Counter()
+            for plist in all_pairs:
+                for w, n in plist:
+                    assert w not in got, "key owned by two ranks"
+                    got[w] = n
+            exp = collections.Counter()
+            for t in all_texts:
+                exp.update(t.split())
+            assert got == exp
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_wordcount_job_gloo_ws2(tmp_path):
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    torch.multiprocessing.spawn(
+        _dist_worker, args=(2, port, str(tmp_path)), nprocs=2, join=True)

@@ -1,0 +1,232 @@
+"""Numerics tests for the CDNA4 HIP kernels vs plain CPU references.
+
+Every kernel is compared against a NumPy/PyTorch CPU oracle on random data
+(guide rule: asymmetric, non-trivial inputs)."""
+
+import collections
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def dev():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from mapreduce_amd import ops
+    ops.require_gpu_ext()
+    return torch.device("cuda:0")
+
+
+def u64view(t_i64_cpu):
+    return t_i64_cpu.numpy().view(np.uint64)
+
+
+TEXT = (b"the quick brown fox jumps over the lazy dog\n"
+        b"pack my box  with five dozen liquor jugs\t\n"
+        b"the the the end x yz  multi   spaces\r\n" * 50)
+
+
+def py_tokenize(data: bytes):
+    from mapreduce_amd.utils.tuple import fnv1a64
+    words = data.split()
+    return [fnv1a64(w) for w in words], words
+
+
+def test_tokenize_matches_python_split(dev):
+    from mapreduce_amd import ops
+    text = torch.frombuffer(bytearray(TEXT), dtype=torch.uint8).to(dev)
+    h, p, n = ops.tokenize_words(text)
+    exp_hashes, exp_words = py_tokenize(TEXT)
+    assert n == len(exp_words)
+    got = sorted(u64view(h.cpu()).tolist())
+    assert got == sorted(exp_hashes)
+
+
+def test_tokenize_edge_cases(dev):
+    from mapreduce_amd import ops
+    for data in (b"", b"   ", b"a", b" a", b"a ", b"ab\ncd", b"\t\n x \r\n",
+                 b"x" * 70000):
+        text = torch.frombuffer(bytearray(data or b"\x20"),
+                                dtype=torch.uint8).to(dev)
+        if data == b"":
+            text = text[:0]
+        h, p, n = ops.tokenize_words(text)
+        exp_hashes, exp_words = py_tokenize(data)
+        assert n == len(exp_words), data
+        assert sorted(u64view(h.cpu()).tolist()) == sorted(exp_hashes), data
+
+
+def test_hash_table_count_vs_counter(dev):
+    from mapreduce_amd import ops
+    rng = np.random.default_rng(42)
+    # zipf-ish duplicates
+    keys_np = rng.integers(0, 5000, size=200_000, dtype=np.uint64) ** 3 + 7
+    keys = torch.from_numpy(keys_np.view(np.int64)).to(dev)
+    ht = ops.HashTable(20_000, dev, exemplar=False)
+    ht.insert_count(keys, None)
+    uk, uv, _ = ht.extract()
+    got = dict(zip(u64view(uk.cpu()).tolist(), uv.cpu().tolist()))
+    exp = collections.Counter(keys_np.tolist())
+    assert got == dict(exp)
+
+
+def test_hash_table_sum_vs_numpy(dev):
+    from mapreduce_amd import ops
+    rng = np.random.default_rng(1)
+    keys_np = rng.integers(0, 1000, size=50_000, dtype=np.uint64) * 2654435761
+    vals_np = rng.integers(-100, 100, size=50_000, dtype=np.int64)
+    keys = torch.from_numpy(keys_np.view(np.int64)).to(dev)
+    vals = torch.from_numpy(vals_np).to(dev)
+    ht = ops.HashTable(2000, dev, exemplar=False)
+    ht.insert_sum(keys, vals)
+    uk, uv, _ = ht.extract()
+    got = dict(zip(u64view(uk.cpu()).tolist(), uv.cpu().tolist()))
+    exp = {}
+    for k, v in zip(keys_np.tolist(), vals_np.tolist()):
+        exp[k] = exp.get(k, 0) + v
+    assert got == exp
+
+
+@pytest.mark.parametrize("n", [0, 1, 63, 64, 2048, 2049, 1_000_000])
+def test_radix_sort_keys(dev, n):
+    from mapreduce_amd import ops
+    rng = np.random.default_rng(n + 1)
+    keys_np = rng.integers(0, 2 ** 63 - 1, size=n, dtype=np.uint64)
+    keys_np |= rng.integers(0, 2, size=n, dtype=np.uint64) << 63  # top bit too
+    keys = torch.from_numpy(keys_np.view(np.int64)).to(dev)
+    (sk,) = ops.sort_by_key(keys)
+    got = u64view(sk.cpu())
+    assert np.array_equal(got, np.sort(keys_np))
+
+
+def test_radix_sort_pairs_stable(dev):
+    from mapreduce_amd import ops
+    rng = np.random.default_rng(3)
+    n = 300_000
+    keys_np = rng.integers(0, 64, size=n, dtype=np.uint64)  # heavy dupes
+    vals_np = np.arange(n, dtype=np.uint64)
+    keys = torch.from_numpy(keys_np.view(np.int64)).to(dev)
+    vals = torch.from_numpy(vals_np.view(np.int64)).to(dev)
+    sk, sv = ops.sort_pairs(keys, vals)
+    got_k = u64view(sk.cpu())
+    got_v = u64view(sv.cpu())
+    order = np.argsort(keys_np, kind="stable")
+    assert np.array_equal(got_k, keys_np[order])
+    assert np.array_equal(got_v, vals_np[order])  # stability
+
+
+def test_radix_sort_fewer_bits(dev):
+    from mapreduce_amd import ops
+    rng = np.random.default_rng(5)
+    keys_np = rng.integers(0, 2 ** 16, size=100_000, dtype=np.uint64)
+    keys = torch.from_numpy(keys_np.view(np.int64)).to(dev)
+    (sk,) = ops.sort_by_key(keys, bits=16)
+    assert np.array_equal(u64view(sk.cpu()), np.sort(keys_np))
+
+
+def test_reduce_by_key_sorted_i64(dev):
+    from mapreduce_amd import ops
+    rng = np.random.default_rng(7)
+    n = 400_000
+    keys_np = np.sort(rng.integers(0, 10_000, size=n, dtype=np.uint64) * 7919)
+    vals_np = rng.integers(0, 1000, size=n, dtype=np.int64)
+    keys = torch.from_numpy(keys_np.view(np.int64)).to(dev)
+    vals = torch.from_numpy(vals_np).to(dev)
+    uk, uv, _, nseg = ops.reduce_by_key_sorted(keys, vals)
+    exp_keys, idx = np.unique(keys_np, return_index=True)
+    exp_sums = np.add.reduceat(vals_np, idx)
+    assert nseg == len(exp_keys)
+    assert np.array_equal(u64view(uk.cpu()), exp_keys)
+    assert np.array_equal(uv.cpu().numpy(), exp_sums)
+
+
+def test_reduce_by_key_counts_and_aux(dev):
+    from mapreduce_amd import ops
+    keys_np = np.sort(np.repeat(
+        np.array([5, 9, 9, 9], dtype=np.uint64) * 10 ** 15, [3, 1, 2, 1]))
+    aux_np = np.arange(len(keys_np), dtype=np.uint64) + 100
+    keys = torch.from_numpy(keys_np.view(np.int64)).to(dev)
+    aux = torch.from_numpy(aux_np.view(np.int64)).to(dev)
+    uk, uv, ua, nseg = ops.reduce_by_key_sorted(keys, None, aux)
+    assert nseg == 2
+    assert uv.cpu().tolist() == [3, 4]
+    assert u64view(ua.cpu()).tolist() == [100, 103]  # first aux per segment
+
+
+def test_reduce_by_key_f64(dev):
+    from mapreduce_amd import ops
+    rng = np.random.default_rng(11)
+    n = 100_000
+    keys_np = np.sort(rng.integers(0, 500, size=n, dtype=np.uint64))
+    vals_np = rng.standard_normal(n)
+    keys = torch.from_numpy(keys_np.view(np.int64)).to(dev)
+    vals = torch.from_numpy(vals_np).to(dev)
+    uk, uv, _, nseg = ops.reduce_by_key_sorted(keys, vals)
+    exp_keys, idx = np.unique(keys_np, return_index=True)
+    exp = np.add.reduceat(vals_np, idx)
+    assert np.allclose(uv.cpu().numpy(), exp, rtol=1e-12, atol=1e-9)
+
+
+def test_partition_hist_matches_mulhi(dev):
+    from mapreduce_amd import ops
+    rng = np.random.default_rng(13)
+    n = 500_000
+    keys_np = rng.integers(0, 2 ** 64 - 1, size=n, dtype=np.uint64)
+    keys = torch.from_numpy(keys_np.view(np.int64)).to(dev)
+    for P in (1, 2, 7, 8, 15, 64):
+        hist = ops.partition_counts(keys, P).cpu().numpy()
+        exp = np.bincount(
+            ((keys_np.astype(object) * P) >> 64).astype(np.int64),
+            minlength=P)
+        assert np.array_equal(hist, exp), P
+        assert hist.sum() == n
+
+
+def test_extract_words_roundtrip(dev):
+    from mapreduce_amd import ops
+    data = b"alpha beta gamma  delta\nepsilon"
+    text = torch.frombuffer(bytearray(data), dtype=torch.uint8).to(dev)
+    h, p, n = ops.tokenize_words(text)
+    lens, blob = ops.extract_words(text, p)
+    words = []
+    raw = bytes(blob.cpu().numpy().tobytes())
+    off = 0
+    for L in lens.cpu().tolist():
+        words.append(raw[off:off + L])
+        off += L
+    assert sorted(words) == sorted(data.split())
+
+
+def test_gpu_wordcount_pipeline_vs_counter(dev):
+    """Fused single-GPU wordcount: tokenize -> hash combine -> sort uniques
+    -> counts, vs collections.Counter (the naive oracle)."""
+    from mapreduce_amd import ops
+    rng = np.random.default_rng(17)
+    vocab = [f"w{i}".encode() for i in range(2000)]
+    words = rng.choice(len(vocab), size=300_000,
+                       p=np.arange(len(vocab), 0, -1) /
+                       np.arange(len(vocab), 0, -1).sum())
+    data = b" ".join(vocab[i] for i in words.tolist())
+    text = torch.frombuffer(bytearray(data), dtype=torch.uint8).to(dev)
+    h, p, n = ops.tokenize_words(text)
+    assert n == len(words)
+    ht = ops.HashTable(4000, dev)
+    ht.insert_count(h, p)
+    uk, uv, up = ht.extract()
+    sk, sv, sp = ops.sort_by_key(uk, uv, up)
+    lens, blob = ops.extract_words(text, sp)
+    raw = bytes(blob.cpu().numpy().tobytes())
+    got = {}
+    off = 0
+    for L, c in zip(lens.cpu().tolist(), sv.cpu().tolist()):
+        got[raw[off:off + L]] = c
+        off += L
+    exp = collections.Counter(vocab[i] for i in words.tolist())
+    assert got == dict(exp)
+    # sortedness of the final key order (u64 bit order)
+    ks = u64view(sk.cpu())
+    assert np.array_equal(ks, np.sort(ks))
