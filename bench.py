@@ -37,6 +37,8 @@ def main() -> int:
     p.add_argument("--splits", type=int, default=197)
     p.add_argument("--vocab", type=int, default=130_000)
     p.add_argument("--device", default=None)
+    p.add_argument("--mode", default="auto",
+                   choices=["auto", "streaming", "fused"])
     args = p.parse_args()
 
     from mapreduce_amd import ops
@@ -50,7 +52,8 @@ def main() -> int:
 
     corpus = make_corpus(device, nwords=args.words, nsplits=args.splits,
                          vocab_size=args.vocab, seed=1234 + rank)
-    job = WordCountJob(device, vocab_estimate=max(args.vocab, 1 << 12))
+    job = WordCountJob(device, vocab_estimate=max(args.vocab, 1 << 12),
+                       mode=args.mode)
     splits = corpus.splits()
 
     def sync():

@@ -54,16 +54,46 @@ class WordCountResult:
 
 
 class WordCountJob:
-    def __init__(self, device, vocab_estimate: int = 1 << 18, group=None):
+    """mode:
+      "streaming" (GPU default) — tokenize_spill -> top-byte radix
+        bucketize -> per-bucket LDS count (all per-word atomics in LDS);
+      "fused" — tokenize straight into the global hash table (slower on
+        Zipf text: per-word probes are L2-latency-bound; kept for A/B and
+        as the low-memory path)."""
+
+    def __init__(self, device, vocab_estimate: int = 1 << 18, group=None,
+                 mode: str = "auto"):
         self.device = torch.device(device)
         self.group = group
         self.rank, self.world = dx.world_info(group)
+        self.vocab_estimate = vocab_estimate
+        if mode == "auto":
+            mode = "streaming" if self.device.type == "cuda" else "fused"
+        self.mode = mode
         self.table = ops.make_table(vocab_estimate, self.device)
         self._nwords = torch.zeros(1, dtype=torch.int64, device=self.device)
 
-    def reset(self, vocab_estimate: int = 1 << 18):
-        self.table = ops.make_table(vocab_estimate, self.device)
+    def reset(self, vocab_estimate: int = 0):
+        self.table = ops.make_table(vocab_estimate or self.vocab_estimate,
+                                    self.device)
         self._nwords.zero_()
+
+    def _map_combine_streaming(self, text: torch.Tensor, s0: int, e0: int
+                               ) -> int:
+        """Streaming map+combine: spill (hash,pos), bucketize by top byte
+        (one radix pass), LDS-count each bucket into the global table.
+        Returns the word count (the one host sync of the map phase)."""
+        chunk = text[s0:e0]
+        cap = chunk.numel() // 2 + 16
+        h, p, c = ops.ext().tokenize_spill(chunk, s0, cap)
+        n = int(c.item())
+        h, p = h[:n], p[:n]
+        hk, pv, totals = ops.ext().radix_pass(h, p, 56)
+        bucket_off = torch.zeros(257, dtype=torch.int64, device=self.device)
+        torch.cumsum(totals, 0, out=bucket_off[1:])
+        ops.ext().bucket_count(hk, pv, bucket_off, 256, 8, self.table.tkeys,
+                               self.table.tvals, self.table.texm)
+        return n
 
     @staticmethod
     def _coalesced(text: torch.Tensor, splits) -> bool:
@@ -90,7 +120,11 @@ class WordCountJob:
         # launches serialize (measured 98% of step time before fusing).
         # Split boundaries are whitespace-aligned, so tokenization over the
         # coalesced range is byte-identical to per-split runs.
-        if self._coalesced(text, splits):
+        nwords_host = None
+        if self.mode == "streaming" and self._coalesced(text, splits):
+            s0, e0 = splits[0][0], splits[-1][1]
+            nwords_host = self._map_combine_streaming(text, s0, e0)
+        elif self._coalesced(text, splits):
             s0, e0 = splits[0][0], splits[-1][1]
             self.table.tokenize_count(text[s0:e0], s0, self._nwords)
         else:
@@ -137,6 +171,7 @@ class WordCountJob:
             fk, fv, fp = sk, sv, sp
             blob_src = text
 
-        nwords = int(self._nwords.item())
+        nwords = (nwords_host if nwords_host is not None
+                  else int(self._nwords.item()))
         return WordCountResult(keys=fk, counts=fv, pos=fp,
                                blob_src=blob_src, nwords=nwords)

@@ -202,6 +202,155 @@ __global__ __launch_bounds__(256) void tokenize_count_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// K2 streaming form: tokenize -> compact (hash, pos) spill arrays.
+// One global offset reservation per THREAD per tile (not per word), writes
+// mostly contiguous.  Feeds the bucketize + LDS-count pipeline below — the
+// streaming replacement for per-word random table probes (which measured
+// latency-bound at ~600 cycles/word).
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void tokenize_spill_kernel(
+    const u8* __restrict__ text, long n, u64 pos_base,
+    u64* __restrict__ out_hash, u64* __restrict__ out_pos,
+    unsigned long long* __restrict__ counter, long cap) {
+  __shared__ u8 tile[TOK_TILE + TOK_HALO];
+  long tile0 = (long)blockIdx.x * TOK_TILE;
+  long tstride = (long)gridDim.x * TOK_TILE;
+  for (long base = tile0; base < n; base += tstride) {
+    __syncthreads();
+    long avail = n - base;
+    long want = avail < TOK_TILE + TOK_HALO ? avail : TOK_TILE + TOK_HALO;
+    for (int o = threadIdx.x * 16; o < want; o += blockDim.x * 16) {
+      if (o + 16 <= want && (((uintptr_t)&text[base + o]) & 15) == 0) {
+        *(uint4*)&tile[o] = *(const uint4*)&text[base + o];
+      } else {
+        for (int b = 0; b < 16 && o + b < want; ++b)
+          tile[o + b] = text[base + o + b];
+      }
+    }
+    __syncthreads();
+    long my0 = (long)threadIdx.x * TOK_BYTES;
+    long myend = my0 + TOK_BYTES;
+    if (myend > avail) myend = avail;
+    if (my0 >= myend) continue;
+    // first pass: find words in my window (register-buffered, <= 8/window)
+    u64 wh[8];
+    u64 wp[8];
+    int nw = 0;
+    u8 prev = (base + my0 == 0) ? ' ' : (my0 ? tile[my0 - 1] : text[base - 1]);
+    for (long i = my0; i < myend; ++i) {
+      u8 c = tile[i];
+      if (!is_ws(c) && is_ws(prev)) {
+        u64 h = FNV64_OFFSET;
+        long j = i;
+        while (j < want) {
+          u8 cc = tile[j];
+          if (is_ws(cc)) break;
+          h ^= cc;
+          h *= FNV64_PRIME;
+          ++j;
+        }
+        if (j == want && base + j < n) {
+          long g = base + j;
+          while (g < n) {
+            u8 cc = text[g];
+            if (is_ws(cc)) break;
+            h ^= cc;
+            h *= FNV64_PRIME;
+            ++g;
+          }
+          j = g - base;
+        }
+        long len = j - i;
+        if (len > 0xFFFF) len = 0xFFFF;
+        wh[nw] = remap_key(h);  // keep HT_EMPTY free for table sentinels
+        wp[nw] = ((pos_base + (u64)(base + i)) << 16) | (u64)len;
+        ++nw;
+      }
+      prev = c;
+    }
+    // one reservation per thread (wave-coalesced by the compiler where
+    // possible), then contiguous writes
+    if (nw) {
+      unsigned long long o = atomicAdd(counter, (unsigned long long)nw);
+      for (int w = 0; w < nw; ++w) {
+        if ((long)o + w < cap) {
+          out_hash[o + w] = wh[w];
+          out_pos[o + w] = wp[w];
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// K5 bucketized count: input (hash, pos) grouped by top-8-bit bucket (one
+// radix partition pass).  grid = NBUCKETS x SLICES blocks; each block
+// LDS-counts its slice of one bucket — a bucket's distinct keys (~vocab/256)
+// fit the 2048-slot LDS table, so every per-word atomic is an LDS atomic;
+// the global table sees only per-block flushes of distinct keys.
+// ---------------------------------------------------------------------------
+
+#define BKT_SLOTS 2048  // LDS table slots (power of 2)
+
+__global__ __launch_bounds__(256) void bucket_count_kernel(
+    const u64* __restrict__ hashes, const u64* __restrict__ pos,
+    const i64* __restrict__ bucket_off,  // [nbuckets+1] exclusive offsets
+    int nbuckets, int slices, u64* __restrict__ tkeys,
+    i64* __restrict__ tvals, u64* __restrict__ texm, u64 cap_mask) {
+  __shared__ u64 ckeys[BKT_SLOTS];
+  __shared__ u64 cpos[BKT_SLOTS];
+  __shared__ u32 ccnt[BKT_SLOTS];
+  int bucket = blockIdx.x / slices;
+  int slice = blockIdx.x % slices;
+  if (bucket >= nbuckets) return;
+  long b0 = bucket_off[bucket];
+  long b1 = bucket_off[bucket + 1];
+  long bn = b1 - b0;
+  long per = (bn + slices - 1) / slices;
+  long s0 = b0 + (long)slice * per;
+  long s1 = s0 + per < b1 ? s0 + per : b1;
+  if (s0 >= s1) return;
+  for (int s = threadIdx.x; s < BKT_SLOTS; s += blockDim.x) {
+    ckeys[s] = HT_EMPTY;
+    ccnt[s] = 0;
+  }
+  __syncthreads();
+  for (long i = s0 + threadIdx.x; i < s1; i += blockDim.x) {
+    u64 k = hashes[i];
+    u64 p = pos[i];
+    u32 slot = (u32)((k ^ (k >> 17)) & (BKT_SLOTS - 1));
+    bool done = false;
+    for (int pr = 0; pr < 64; ++pr) {
+      u64 cur = ckeys[slot];
+      if (cur == k) {
+        atomicAdd(&ccnt[slot], 1u);
+        done = true;
+        break;
+      }
+      if (cur == HT_EMPTY) {
+        u64 prevk = atomicCAS((unsigned long long*)&ckeys[slot],
+                              (unsigned long long)HT_EMPTY,
+                              (unsigned long long)k);
+        if (prevk == HT_EMPTY) cpos[slot] = p;
+        if (prevk == HT_EMPTY || prevk == k) {
+          atomicAdd(&ccnt[slot], 1u);
+          done = true;
+          break;
+        }
+      }
+      slot = (slot + 1) & (BKT_SLOTS - 1);
+    }
+    if (!done)  // pathological bucket: spill straight to the global table
+      ht_add(k, p, 1, tkeys, tvals, texm, cap_mask);
+  }
+  __syncthreads();
+  for (int s = threadIdx.x; s < BKT_SLOTS; s += blockDim.x)
+    if (ckeys[s] != HT_EMPTY && ccnt[s])
+      ht_add(ckeys[s], cpos[s], (i64)ccnt[s], tkeys, tvals, texm, cap_mask);
+}
+
+// ---------------------------------------------------------------------------
 // K5 (aggregation form): open-addressing hash table, linear probing.
 // Combiner for declared associative+commutative reducers (job.lua:104-106 —
 // the reference's own fast-path flags select this path).

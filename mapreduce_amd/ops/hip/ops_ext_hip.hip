@@ -75,6 +75,41 @@ void tokenize_count(torch::Tensor text, long pos_base, torch::Tensor tkeys,
                      reinterpret_cast<unsigned long long*>(nwords.data_ptr<i64>()));
 }
 
+// ------------------------------------------------------------- K2 streaming
+std::vector<torch::Tensor> tokenize_spill(torch::Tensor text, long pos_base,
+                                          long cap) {
+  TORCH_CHECK(text.is_cuda() && text.scalar_type() == torch::kUInt8 &&
+              text.is_contiguous(), "text must be contiguous u8 on GPU");
+  long n = text.numel();
+  auto opts = torch::TensorOptions().device(text.device()).dtype(torch::kInt64);
+  auto out_hash = torch::empty({cap}, opts);
+  auto out_pos = torch::empty({cap}, opts);
+  auto counter = torch::zeros({1}, opts);
+  if (n)
+    hipLaunchKernelGGL(tokenize_spill_kernel, dim3(grid_for(n, TOK_BYTES)),
+                       dim3(kBlock), 0, cur_stream(), text.data_ptr<u8>(), n,
+                       (u64)pos_base, u64p(out_hash), u64p(out_pos),
+                       reinterpret_cast<unsigned long long*>(counter.data_ptr<i64>()),
+                       cap);
+  return {out_hash, out_pos, counter};
+}
+
+// ---------------------------------------------------------- K5 bucket count
+void bucket_count(torch::Tensor hashes, torch::Tensor pos,
+                  torch::Tensor bucket_off, long nbuckets, long slices,
+                  torch::Tensor tkeys, torch::Tensor tvals,
+                  torch::Tensor texm) {
+  check_dev_i64(hashes, "hashes");
+  long cap = tkeys.numel();
+  TORCH_CHECK((cap & (cap - 1)) == 0, "table capacity must be a power of 2");
+  TORCH_CHECK(bucket_off.numel() == nbuckets + 1, "bucket_off size");
+  hipLaunchKernelGGL(bucket_count_kernel, dim3(nbuckets * slices),
+                     dim3(kBlock), 0, cur_stream(), u64cp(hashes),
+                     u64cp(pos), bucket_off.data_ptr<i64>(), (int)nbuckets,
+                     (int)slices, u64p(tkeys), tvals.data_ptr<i64>(),
+                     texm.numel() ? u64p(texm) : nullptr, (u64)(cap - 1));
+}
+
 // ----------------------------------------------------------------------- K5a
 void hash_insert_count(torch::Tensor keys, torch::Tensor pos,
                        torch::Tensor tkeys, torch::Tensor tvals,
@@ -207,6 +242,35 @@ torch::Tensor gather_bytes(torch::Tensor text, torch::Tensor pos,
 }
 
 // ------------------------------------------------------------------------ K1
+// single partition pass on one 8-bit digit; returns (keys, vals, per-digit
+// totals) — used to bucketize by top byte before LDS counting
+std::vector<torch::Tensor> radix_pass(torch::Tensor keys, torch::Tensor vals,
+                                      long shift) {
+  check_dev_i64(keys, "keys");
+  long n = keys.numel();
+  bool has_vals = vals.numel() > 0;
+  auto opts = keys.options();
+  long ntiles = (n + RS_TILE - 1) / RS_TILE;
+  if (ntiles == 0) ntiles = 1;
+  auto kout = torch::empty({n}, opts);
+  auto vout = has_vals ? torch::empty({n}, opts) : torch::empty({0}, opts);
+  auto hist = torch::zeros({(long)RS_BINS * ntiles}, opts);
+  if (n) {
+    hipLaunchKernelGGL(radix_hist_kernel, dim3(ntiles), dim3(RS_BLOCK), 0,
+                       cur_stream(), u64cp(keys), n, (int)shift, ntiles,
+                       hist.data_ptr<i64>());
+    auto scanned = torch::cumsum(hist, 0);
+    auto base = scanned - hist;
+    hipLaunchKernelGGL(radix_scatter_kernel, dim3(ntiles), dim3(RS_BLOCK), 0,
+                       cur_stream(), u64cp(keys),
+                       has_vals ? u64cp(vals) : nullptr, n, (int)shift,
+                       ntiles, base.data_ptr<i64>(), u64p(kout),
+                       has_vals ? u64p(vout) : nullptr);
+  }
+  auto totals = hist.view({(long)RS_BINS, ntiles}).sum(1);
+  return {kout, vout, totals};
+}
+
 std::vector<torch::Tensor> radix_sort_pairs(torch::Tensor keys,
                                             torch::Tensor vals, int bits) {
   check_dev_i64(keys, "keys");
@@ -245,6 +309,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("tokenize", &tokenize, "tokenize text -> (hash, pos, count)");
   m.def("tokenize_count", &tokenize_count,
         "fused tokenize + hash-table count");
+  m.def("tokenize_spill", &tokenize_spill,
+        "tokenize -> compact (hash,pos) arrays");
+  m.def("bucket_count", &bucket_count,
+        "LDS count of bucket-partitioned (hash,pos)");
   m.def("hash_insert_count", &hash_insert_count);
   m.def("hash_insert_sum_i64", &hash_insert_sum_i64);
   m.def("hash_extract", &hash_extract);
@@ -256,4 +324,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pos_len", &pos_len);
   m.def("gather_bytes", &gather_bytes);
   m.def("radix_sort_pairs", &radix_sort_pairs);
+  m.def("radix_pass", &radix_pass);
 }
