@@ -80,19 +80,28 @@ class WordCountJob:
 
     def _map_combine_streaming(self, text: torch.Tensor, s0: int, e0: int
                                ) -> int:
-        """Streaming map+combine: spill (hash,pos), bucketize by top byte
-        (one radix pass), LDS-count each bucket into the global table.
-        Returns the word count (the one host sync of the map phase)."""
+        """Streaming map+combine: tokenize with per-block LDS caches (Zipf
+        head aggregates in LDS); cache misses spill as (hash,pos), get
+        bucketized by top byte (one radix pass) and LDS-counted per bucket.
+        No per-word global-memory probe on any path.  Returns the word
+        count (the one host sync of the map phase)."""
         chunk = text[s0:e0]
         cap = chunk.numel() // 2 + 16
-        h, p, c = ops.ext().tokenize_spill(chunk, s0, cap)
-        n = int(c.item())
-        h, p = h[:n], p[:n]
-        hk, pv, totals = ops.ext().radix_pass(h, p, 56)
-        bucket_off = torch.zeros(257, dtype=torch.int64, device=self.device)
-        torch.cumsum(totals, 0, out=bucket_off[1:])
-        ops.ext().bucket_count(hk, pv, bucket_off, 256, 8, self.table.tkeys,
-                               self.table.tvals, self.table.texm)
+        h, p, c = ops.ext().tokenize_cache_spill(
+            chunk, s0, self.table.tkeys, self.table.tvals, self.table.texm,
+            cap, self._nwords)
+        cnt = c.item()          # spill count (syncs; nwords rides along)
+        n = int(self._nwords.item())
+        nspill = int(cnt)
+        if nspill:
+            h, p = h[:nspill], p[:nspill]
+            hk, pv, totals = ops.ext().radix_pass(h, p, 56)
+            bucket_off = torch.zeros(257, dtype=torch.int64,
+                                     device=self.device)
+            torch.cumsum(totals, 0, out=bucket_off[1:])
+            ops.ext().bucket_count(hk, pv, bucket_off, 256, 8,
+                                   self.table.tkeys, self.table.tvals,
+                                   self.table.texm)
         return n
 
     @staticmethod

@@ -94,6 +94,33 @@ std::vector<torch::Tensor> tokenize_spill(torch::Tensor text, long pos_base,
   return {out_hash, out_pos, counter};
 }
 
+// -------------------------------------------------------------- K2+K5 v4
+std::vector<torch::Tensor> tokenize_cache_spill(
+    torch::Tensor text, long pos_base, torch::Tensor tkeys,
+    torch::Tensor tvals, torch::Tensor texm, long spill_cap,
+    torch::Tensor nwords) {
+  TORCH_CHECK(text.is_cuda() && text.scalar_type() == torch::kUInt8 &&
+              text.is_contiguous(), "text must be contiguous u8 on GPU");
+  long n = text.numel();
+  long cap = tkeys.numel();
+  TORCH_CHECK((cap & (cap - 1)) == 0, "table capacity must be a power of 2");
+  auto opts = torch::TensorOptions().device(text.device()).dtype(torch::kInt64);
+  auto out_hash = torch::empty({spill_cap}, opts);
+  auto out_pos = torch::empty({spill_cap}, opts);
+  auto counter = torch::zeros({1}, opts);
+  if (n)
+    hipLaunchKernelGGL(tokenize_cache_spill_kernel,
+                       dim3(grid_for(n, TOK_BYTES)), dim3(kBlock), 0,
+                       cur_stream(), text.data_ptr<u8>(), n, (u64)pos_base,
+                       u64p(tkeys), tvals.data_ptr<i64>(),
+                       texm.numel() ? u64p(texm) : nullptr, (u64)(cap - 1),
+                       u64p(out_hash), u64p(out_pos),
+                       reinterpret_cast<unsigned long long*>(counter.data_ptr<i64>()),
+                       spill_cap,
+                       reinterpret_cast<unsigned long long*>(nwords.data_ptr<i64>()));
+  return {out_hash, out_pos, counter};
+}
+
 // ---------------------------------------------------------- K5 bucket count
 void bucket_count(torch::Tensor hashes, torch::Tensor pos,
                   torch::Tensor bucket_off, long nbuckets, long slices,
@@ -311,6 +338,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused tokenize + hash-table count");
   m.def("tokenize_spill", &tokenize_spill,
         "tokenize -> compact (hash,pos) arrays");
+  m.def("tokenize_cache_spill", &tokenize_cache_spill,
+        "tokenize; LDS cache counts the head, misses spill");
   m.def("bucket_count", &bucket_count,
         "LDS count of bucket-partitioned (hash,pos)");
   m.def("hash_insert_count", &hash_insert_count);
