@@ -405,6 +405,132 @@ __global__ __launch_bounds__(256) void tokenize_cache_spill_kernel(
   if (lane_id() == 0 && ws) atomicAdd(nwords, ws);
 }
 
+// Ablation copy of the tokenize kernel (diagnosis only — §5.4 rule 8:
+// ablate empirically before optimizing).  mode: 1=stage tiles only,
+// 2=+word-boundary scan (count only), 3=+FNV hash, 4=+LDS cache insert,
+// 5=full (with spill).  Results accumulate into sink to stay live.
+__global__ __launch_bounds__(256) void tok_ablate_kernel(
+    const u8* __restrict__ text, long n, int mode,
+    u64* __restrict__ ckeys_g /*scratch cap>=2^20*/, i64* __restrict__ sink,
+    u64* __restrict__ out_hash, u64* __restrict__ out_pos,
+    unsigned long long* __restrict__ spill_counter, long spill_cap) {
+  __shared__ u8 tile[TOK_TILE + TOK_HALO];
+  __shared__ u64 ckeys[TOK_CACHE];
+  __shared__ u64 cpos[TOK_CACHE];
+  __shared__ u32 ccnt[TOK_CACHE];
+  for (int s = threadIdx.x; s < TOK_CACHE; s += blockDim.x) {
+    ckeys[s] = HT_EMPTY;
+    ccnt[s] = 0;
+  }
+  unsigned long long acc = 0;
+  long tile0 = (long)blockIdx.x * TOK_TILE;
+  long tstride = (long)gridDim.x * TOK_TILE;
+  for (long base = tile0; base < n; base += tstride) {
+    __syncthreads();
+    long avail = n - base;
+    long want = avail < TOK_TILE + TOK_HALO ? avail : TOK_TILE + TOK_HALO;
+    for (int o = threadIdx.x * 16; o < want; o += blockDim.x * 16) {
+      if (o + 16 <= want && (((uintptr_t)&text[base + o]) & 15) == 0) {
+        *(uint4*)&tile[o] = *(const uint4*)&text[base + o];
+      } else {
+        for (int b = 0; b < 16 && o + b < want; ++b)
+          tile[o + b] = text[base + o + b];
+      }
+    }
+    __syncthreads();
+    if (mode == 1) {  // consume a few tile bytes so staging stays live
+      acc += tile[threadIdx.x];
+      continue;
+    }
+    long my0 = (long)threadIdx.x * TOK_BYTES;
+    long myend = my0 + TOK_BYTES;
+    if (myend > avail) myend = avail;
+    if (my0 >= myend) continue;
+    u64 sh[8];
+    u64 sp[8];
+    int ns = 0;
+    u8 prev = (base + my0 == 0) ? ' ' : (my0 ? tile[my0 - 1] : text[base - 1]);
+    for (long i = my0; i < myend; ++i) {
+      u8 c = tile[i];
+      if (!is_ws(c) && is_ws(prev)) {
+        u64 h = FNV64_OFFSET;
+        long j = i;
+        if (mode == 2) {  // boundary-scan only: find end without hashing
+          while (j < want && !is_ws(tile[j])) ++j;
+          acc += (u64)(j - i);
+        } else {
+          while (j < want) {
+            u8 cc = tile[j];
+            if (is_ws(cc)) break;
+            h ^= cc;
+            h *= FNV64_PRIME;
+            ++j;
+          }
+          if (j == want && base + j < n) {
+            long g = base + j;
+            while (g < n && !is_ws(text[g])) {
+              h ^= text[g];
+              h *= FNV64_PRIME;
+              ++g;
+            }
+            j = g - base;
+          }
+        }
+        if (mode == 3) {
+          acc += h;
+        } else if (mode >= 4) {
+          u64 k = remap_key(h);
+          u64 p = ((u64)(base + i) << 16) | (u64)(j - i);
+          u32 slot = (u32)((k ^ (k >> 32)) & (TOK_CACHE - 1));
+          bool done = false;
+          for (int pr = 0; pr < TOK_PROBE; ++pr) {
+            u64 cur = ckeys[slot];
+            if (cur == k) {
+              atomicAdd(&ccnt[slot], 1u);
+              done = true;
+              break;
+            }
+            if (cur == HT_EMPTY) {
+              u64 prevk = atomicCAS((unsigned long long*)&ckeys[slot],
+                                    (unsigned long long)HT_EMPTY,
+                                    (unsigned long long)k);
+              if (prevk == HT_EMPTY) cpos[slot] = p;
+              if (prevk == HT_EMPTY || prevk == k) {
+                atomicAdd(&ccnt[slot], 1u);
+                done = true;
+                break;
+              }
+            }
+            slot = (slot + 1) & (TOK_CACHE - 1);
+          }
+          if (!done) {
+            if (mode == 4) acc += k;
+            else {
+              sh[ns] = k;
+              sp[ns] = p;
+              ++ns;
+            }
+          }
+        }
+      }
+      prev = c;
+    }
+    if (mode >= 5 && ns) {
+      unsigned long long o = atomicAdd(spill_counter, (unsigned long long)ns);
+      for (int w = 0; w < ns; ++w)
+        if ((long)o + w < spill_cap) {
+          out_hash[o + w] = sh[w];
+          out_pos[o + w] = sp[w];
+        }
+    }
+  }
+  __syncthreads();
+  if (mode >= 4)
+    for (int s = threadIdx.x; s < TOK_CACHE; s += blockDim.x)
+      acc += ccnt[s];
+  if (acc) atomicAdd((unsigned long long*)sink, acc);
+}
+
 // ---------------------------------------------------------------------------
 // K5 bucketized count: input (hash, pos) grouped by top-8-bit bucket (one
 // radix partition pass).  grid = NBUCKETS x SLICES blocks; each block

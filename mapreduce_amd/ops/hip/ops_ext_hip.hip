@@ -121,6 +121,44 @@ std::vector<torch::Tensor> tokenize_cache_spill(
   return {out_hash, out_pos, counter};
 }
 
+// diagnosis only
+double tok_ablate(torch::Tensor text, long mode, long iters) {
+  long n = text.numel();
+  auto opts = torch::TensorOptions().device(text.device()).dtype(torch::kInt64);
+  auto sink = torch::zeros({1}, opts);
+  long cap = n / 2 + 16;
+  auto oh = torch::empty({mode >= 5 ? cap : 1}, opts);
+  auto op = torch::empty({mode >= 5 ? cap : 1}, opts);
+  auto ctr = torch::zeros({1}, opts);
+  hipStream_t st = cur_stream();
+  hipEvent_t e0, e1;
+  hipEventCreate(&e0);
+  hipEventCreate(&e1);
+  // warm
+  hipLaunchKernelGGL(tok_ablate_kernel, dim3(grid_for(n, TOK_BYTES)),
+                     dim3(kBlock), 0, st, text.data_ptr<u8>(), n, (int)mode,
+                     nullptr, sink.data_ptr<i64>(), u64p(oh), u64p(op),
+                     reinterpret_cast<unsigned long long*>(ctr.data_ptr<i64>()),
+                     cap);
+  hipEventRecord(e0, st);
+  for (long i = 0; i < iters; ++i) {
+    ctr.zero_();
+    hipLaunchKernelGGL(tok_ablate_kernel, dim3(grid_for(n, TOK_BYTES)),
+                       dim3(kBlock), 0, st, text.data_ptr<u8>(), n,
+                       (int)mode, nullptr, sink.data_ptr<i64>(), u64p(oh),
+                       u64p(op),
+                       reinterpret_cast<unsigned long long*>(ctr.data_ptr<i64>()),
+                       cap);
+  }
+  hipEventRecord(e1, st);
+  hipEventSynchronize(e1);
+  float ms = 0;
+  hipEventElapsedTime(&ms, e0, e1);
+  hipEventDestroy(e0);
+  hipEventDestroy(e1);
+  return (double)ms / iters;
+}
+
 // ---------------------------------------------------------- K5 bucket count
 void bucket_count(torch::Tensor hashes, torch::Tensor pos,
                   torch::Tensor bucket_off, long nbuckets, long slices,
@@ -340,6 +378,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "tokenize -> compact (hash,pos) arrays");
   m.def("tokenize_cache_spill", &tokenize_cache_spill,
         "tokenize; LDS cache counts the head, misses spill");
+  m.def("tok_ablate", &tok_ablate, "ablation timing (diagnosis)");
   m.def("bucket_count", &bucket_count,
         "LDS count of bucket-partitioned (hash,pos)");
   m.def("hash_insert_count", &hash_insert_count);
