@@ -408,6 +408,228 @@ __global__ __launch_bounds__(256) void tokenize_cache_spill_kernel(
   if (lane_id() == 0 && ws) atomicAdd(nwords, ws);
 }
 
+// ---------------------------------------------------------------------------
+// K2+K5 v5: mask-based tokenizer with per-tile word list.
+// The v4 structure paid the cache-insert cost at nearly every byte position
+// (insert code inside the divergent per-byte loop).  v5 decouples:
+//   A. branchless classification — each thread turns its 16 LDS bytes into
+//      a whitespace bitmask (compile-time unrolled extracts, no branches);
+//   B. word enumeration from mask pairs (starts = ~m & (m<<1); length =
+//      ctz of the shifted mask) into a per-tile LDS word list, slots
+//      reserved with one wave-prefix + one LDS atomic per wave;
+//   C. balanced processing — lanes take words round-robin from the list
+//      (every lane active), hash via aligned u64 funnel reads from LDS,
+//      insert into the direct-probed LDS cache, wave-aggregated spill of
+//      misses (one global atomic per wave per round).
+// ---------------------------------------------------------------------------
+
+#define TOKV5_MARK 0xFFFu  // len field marker: word longer than the mask
+                           // window — rescan from global text
+
+__global__ __launch_bounds__(256) void tokenize_v5_kernel(
+    const u8* __restrict__ text, long n, u64 pos_base,
+    u64* __restrict__ tkeys, i64* __restrict__ tvals, u64* __restrict__ texm,
+    u64 cap_mask, u64* __restrict__ out_hash, u64* __restrict__ out_pos,
+    unsigned long long* __restrict__ spill_counter, long spill_cap,
+    unsigned long long* __restrict__ nwords) {
+  __shared__ __align__(16) u8 tile[TOK_TILE + TOK_HALO];
+  __shared__ unsigned short wsmask[260];  // 256 windows + 4 halo windows
+  __shared__ u32 wlist[TOK_TILE / 2 + 64];
+  __shared__ u32 wl_count;
+  __shared__ u64 ckeys[TOK_CACHE];
+  __shared__ u64 cpos[TOK_CACHE];
+  __shared__ u32 ccnt[TOK_CACHE];
+  for (int s = threadIdx.x; s < TOK_CACHE; s += blockDim.x) {
+    ckeys[s] = HT_EMPTY;
+    ccnt[s] = 0;
+  }
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const u64 lt_mask = ((u64)1 << lane) - 1;
+  unsigned long long my_words = 0;
+  long tile0 = (long)blockIdx.x * TOK_TILE;
+  long tstride = (long)gridDim.x * TOK_TILE;
+  for (long base = tile0; base < n; base += tstride) {
+    __syncthreads();  // previous tile fully consumed
+    long avail = n - base;
+    long want = avail < TOK_TILE + TOK_HALO ? avail : TOK_TILE + TOK_HALO;
+    for (int o = tid * 16; o < want; o += blockDim.x * 16) {
+      if (o + 16 <= want && (((uintptr_t)&text[base + o]) & 15) == 0) {
+        *(uint4*)&tile[o] = *(const uint4*)&text[base + o];
+      } else {
+        for (int b = 0; b < 16 && o + b < want; ++b)
+          tile[o + b] = text[base + o + b];
+      }
+    }
+    if (tid == 0) wl_count = 0;
+    __syncthreads();
+    // ---- A: classify 16 bytes -> ws bitmask (bit set = whitespace)
+    {
+      int w0 = tid * 16;
+      u32 m = 0xFFFFu;
+      if (w0 < want) {
+        const uint4 v = *(const uint4*)&tile[w0];
+        m = 0;
+        const u32 rs[4] = {v.x, v.y, v.z, v.w};
+        #pragma unroll
+        for (int r = 0; r < 4; ++r)
+          #pragma unroll
+          for (int b = 0; b < 4; ++b) {
+            u8 c = (u8)(rs[r] >> (8 * b));
+            m |= (u32)is_ws(c) << (r * 4 + b);
+          }
+        long lim = want - w0;
+        if (lim < 16) m |= ~((1u << lim) - 1) & 0xFFFFu;
+      }
+      wsmask[tid] = (unsigned short)m;
+      if (tid < 4) {  // halo windows
+        int h0 = TOK_TILE + tid * 16;
+        u32 hm = 0xFFFFu;
+        if (h0 < want) {
+          hm = 0;
+          #pragma unroll
+          for (int b = 0; b < 16; ++b) {
+            u32 wsb = (h0 + b < want) ? (u32)is_ws(tile[h0 + b]) : 1u;
+            hm |= wsb << b;
+          }
+        }
+        wsmask[256 + tid] = (unsigned short)hm;
+      }
+    }
+    __syncthreads();
+    // ---- B: word starts in my window -> LDS word list
+    {
+      int w0 = tid * 16;
+      u32 m = wsmask[tid];
+      u32 nx = (tid < 255) ? wsmask[tid + 1] : wsmask[256];
+      u32 m32 = m | (nx << 16);
+      u32 prevb;
+      if (tid > 0)
+        prevb = (wsmask[tid - 1] >> 15) & 1u;
+      else
+        prevb = (base == 0) ? 1u : (u32)is_ws(text[base - 1]);
+      u32 sm = ~m32 & ((m32 << 1) | prevb) & 0xFFFFu;
+      // starts must lie inside the data (not the padded region)
+      long lim = avail - w0;
+      if (lim <= 0) sm = 0;
+      else if (lim < 16) sm &= (1u << lim) - 1;
+      int nw = __popc(sm);
+      my_words += nw;
+      // wave-exclusive prefix of nw, one LDS atomic per wave
+      u32 incl = (u32)nw;
+      #pragma unroll
+      for (int off = 1; off < WAVE; off <<= 1) {
+        u32 x = __shfl_up(incl, off, WAVE);
+        if (lane >= off) incl += x;
+      }
+      u32 wave_total = __shfl(incl, WAVE - 1, WAVE);
+      u32 wave_base = 0;
+      if (lane == 0 && wave_total)
+        wave_base = atomicAdd(&wl_count, wave_total);
+      wave_base = __shfl(wave_base, 0, WAVE);
+      u32 slot = wave_base + incl - (u32)nw;
+      while (sm) {
+        int s = __ffs(sm) - 1;
+        sm &= sm - 1;
+        u32 t = m32 >> s;
+        u32 len = t ? (u32)(__ffs(t) - 1) : TOKV5_MARK;
+        if (len == 0) len = TOKV5_MARK;  // cannot happen (bit s clear)
+        wlist[slot++] = (u32)(w0 + s) | (len << 13);
+      }
+    }
+    __syncthreads();
+    // ---- C: balanced hash + count/spill
+    int tot = (int)wl_count;
+    const u64* t64 = (const u64*)tile;
+    for (int widx = tid; widx < tot + (WAVE - 1); widx += blockDim.x) {
+      // padded loop bound keeps whole waves together for ballots; inactive
+      // lanes carry valid=false
+      bool valid = widx < tot;
+      u64 h = FNV64_OFFSET;
+      u64 p = 0;
+      u64 k = 0;
+      if (valid) {
+        u32 pk = wlist[widx];
+        int s = (int)(pk & 0x1FFFu);
+        u32 len = pk >> 13;
+        if (len != TOKV5_MARK) {
+          int q = s >> 3;
+          int sh = (s & 7) * 8;
+          u64 cur = t64[q] >> sh;
+          int have = 8 - (s & 7);
+          for (u32 b = 0; b < len; ++b) {
+            if (have == 0) {
+              cur = t64[++q];
+              have = 8;
+            }
+            h = (h ^ (cur & 0xFF)) * FNV64_PRIME;
+            cur >>= 8;
+            --have;
+          }
+        } else {  // long word: rescan from global
+          long g = base + s;
+          while (g < n && !is_ws(text[g])) {
+            h = (h ^ text[g]) * FNV64_PRIME;
+            ++g;
+          }
+          len = (u32)((g - (base + s)) > 0xFFFF ? 0xFFFF : g - (base + s));
+        }
+        k = remap_key(h);
+        p = ((pos_base + (u64)(base + s)) << 16) | (u64)(len & 0xFFFF);
+      }
+      // cache insert (probe <= TOK_PROBE)
+      bool miss = valid;
+      if (valid) {
+        u32 slot = (u32)((k ^ (k >> 32)) & (TOK_CACHE - 1));
+        for (int pr = 0; pr < TOK_PROBE; ++pr) {
+          u64 cur = ckeys[slot];
+          if (cur == k) {
+            atomicAdd(&ccnt[slot], 1u);
+            miss = false;
+            break;
+          }
+          if (cur == HT_EMPTY) {
+            u64 prevk = atomicCAS((unsigned long long*)&ckeys[slot],
+                                  (unsigned long long)HT_EMPTY,
+                                  (unsigned long long)k);
+            if (prevk == HT_EMPTY) cpos[slot] = p;
+            if (prevk == HT_EMPTY || prevk == k) {
+              atomicAdd(&ccnt[slot], 1u);
+              miss = false;
+              break;
+            }
+          }
+          slot = (slot + 1) & (TOK_CACHE - 1);
+        }
+      }
+      // wave-aggregated spill (one global atomic per wave per round)
+      u64 miss_mask = __ballot(miss);
+      if (miss_mask) {
+        int leader = __ffsll((unsigned long long)miss_mask) - 1;
+        unsigned long long o = 0;
+        if (lane == leader)
+          o = atomicAdd(spill_counter,
+                        (unsigned long long)__popcll(miss_mask));
+        o = __shfl(o, leader, WAVE);
+        if (miss) {
+          long mi = (long)o + __popcll(miss_mask & lt_mask);
+          if (mi < spill_cap) {
+            out_hash[mi] = k;
+            out_pos[mi] = p;
+          }
+        }
+      }
+    }
+  }
+  __syncthreads();
+  for (int s = tid; s < TOK_CACHE; s += blockDim.x)
+    if (ckeys[s] != HT_EMPTY && ccnt[s])
+      ht_add(ckeys[s], cpos[s], (i64)ccnt[s], tkeys, tvals, texm, cap_mask);
+  unsigned long long ws = my_words;
+  for (int off = 32; off > 0; off >>= 1) ws += __shfl_down(ws, off, WAVE);
+  if (lane == 0 && ws) atomicAdd(nwords, ws);
+}
+
 // Ablation copy of the tokenize kernel (diagnosis only — §5.4 rule 8:
 // ablate empirically before optimizing).  mode: 1=stage tiles only,
 // 2=+word-boundary scan (count only), 3=+FNV hash, 4=+LDS cache insert,

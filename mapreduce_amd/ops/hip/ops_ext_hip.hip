@@ -108,16 +108,29 @@ std::vector<torch::Tensor> tokenize_cache_spill(
   auto out_hash = torch::empty({spill_cap}, opts);
   auto out_pos = torch::empty({spill_cap}, opts);
   auto counter = torch::zeros({1}, opts);
-  if (n)
-    hipLaunchKernelGGL(tokenize_cache_spill_kernel,
-                       dim3(grid_for(n, TOK_BYTES)), dim3(kBlock), 0,
-                       cur_stream(), text.data_ptr<u8>(), n, (u64)pos_base,
-                       u64p(tkeys), tvals.data_ptr<i64>(),
-                       texm.numel() ? u64p(texm) : nullptr, (u64)(cap - 1),
-                       u64p(out_hash), u64p(out_pos),
-                       reinterpret_cast<unsigned long long*>(counter.data_ptr<i64>()),
-                       spill_cap,
-                       reinterpret_cast<unsigned long long*>(nwords.data_ptr<i64>()));
+  if (n) {
+    const char* v = getenv("MR_TOKENIZE_V4");  // A/B escape hatch
+    if (v && v[0] == '1')
+      hipLaunchKernelGGL(tokenize_cache_spill_kernel,
+                         dim3(grid_for(n, TOK_BYTES)), dim3(kBlock), 0,
+                         cur_stream(), text.data_ptr<u8>(), n, (u64)pos_base,
+                         u64p(tkeys), tvals.data_ptr<i64>(),
+                         texm.numel() ? u64p(texm) : nullptr, (u64)(cap - 1),
+                         u64p(out_hash), u64p(out_pos),
+                         reinterpret_cast<unsigned long long*>(counter.data_ptr<i64>()),
+                         spill_cap,
+                         reinterpret_cast<unsigned long long*>(nwords.data_ptr<i64>()));
+    else
+      hipLaunchKernelGGL(tokenize_v5_kernel,
+                         dim3(grid_for(n, TOK_BYTES)), dim3(kBlock), 0,
+                         cur_stream(), text.data_ptr<u8>(), n, (u64)pos_base,
+                         u64p(tkeys), tvals.data_ptr<i64>(),
+                         texm.numel() ? u64p(texm) : nullptr, (u64)(cap - 1),
+                         u64p(out_hash), u64p(out_pos),
+                         reinterpret_cast<unsigned long long*>(counter.data_ptr<i64>()),
+                         spill_cap,
+                         reinterpret_cast<unsigned long long*>(nwords.data_ptr<i64>()));
+  }
   return {out_hash, out_pos, counter};
 }
 
