@@ -108,8 +108,13 @@ std::vector<torch::Tensor> tokenize_cache_spill(
   auto out_pos = torch::empty({spill_cap}, opts);
   auto counter = torch::zeros({1}, opts);
   if (n) {
-    const char* v = getenv("MR_TOKENIZE_V4");  // A/B escape hatch
-    if (v && v[0] == '1')
+    // default v4 (cache + spill in the scan loop).  The v5 word-list
+    // restructure measured 2x SLOWER end-to-end (11.5 vs 6.0 ms/step,
+    // same box): its extra barriers, LDS footprint (+12 KB -> fewer
+    // blocks/CU) and per-round ballots outweigh the balance win at this
+    // grain.  Kept behind MR_TOKENIZE_V5=1 as a recorded experiment.
+    const char* v = getenv("MR_TOKENIZE_V5");
+    if (!(v && v[0] == '1'))
       hipLaunchKernelGGL(tokenize_cache_spill_kernel,
                          dim3(grid_for(n, TOK_BYTES)), dim3(kBlock), 0,
                          cur_stream(), text.data_ptr<u8>(), n, (u64)pos_base,
@@ -119,7 +124,7 @@ std::vector<torch::Tensor> tokenize_cache_spill(
                          reinterpret_cast<unsigned long long*>(counter.data_ptr<i64>()),
                          spill_cap,
                          reinterpret_cast<unsigned long long*>(nwords.data_ptr<i64>()));
-    else
+    else  // MR_TOKENIZE_V5=1
       hipLaunchKernelGGL(tokenize_v5_kernel,
                          dim3(grid_for(n, TOK_BYTES)), dim3(kBlock), 0,
                          cur_stream(), text.data_ptr<u8>(), n, (u64)pos_base,
