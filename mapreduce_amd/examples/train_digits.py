@@ -1,0 +1,169 @@
+"""Iterative MapReduce training — the APRIL-ANN example re-expressed.
+
+The reference's MNIST-digits task (examples/APRIL-ANN):
+each iteration, mapfn computes gradients of the current model on its data
+shard and emits (weight_name, gradient); reducefn sums gradients (K6);
+finalfn applies the optimizer step, evaluates, and returns "loop" until
+convergence.  Model state crosses processes via persistent_table (the
+reference serializes the model to GridFS and shares the filename,
+common.lua:24-29, 57-77 — here the state_dict itself rides in the table).
+
+Works on the host tier with torch tensors as emitted values; on a GPU node
+the same reduce maps to the RCCL gradient allreduce (gpu/gradsum.py).
+
+init_args: {"shards": int, "iters": int, "lr": float, "bunch": int,
+"coord_token": optional} — all synthetic data (no network for datasets).
+"""
+
+from __future__ import annotations
+
+import base64
+import io
+import pickle
+
+import torch
+
+_CFG = {"shards": 4, "iters": 3, "lr": 0.1, "bunch": 64, "seed": 0,
+        "hidden": 32, "cnn": None, "db": "mr"}
+STATE = {"model": None, "iteration": 0, "losses": [], "pt": None}
+
+
+def _pt():
+    """Shared-state table for multi-process runs (the reference's
+    persistent_table 'conf' + GridFS model file, common.lua:57-77)."""
+    if STATE["pt"] is None and _CFG["cnn"]:
+        from mapreduce_amd.persistent_table import PersistentTable
+
+        STATE["pt"] = PersistentTable("train_digits", _CFG["cnn"],
+                                      _CFG["db"])
+    return STATE["pt"]
+
+
+def _sync_model_from_pt():
+    """Worker side: refresh the local model replica when the table holds a
+    newer iteration's weights."""
+    pt = _pt()
+    if pt is None:
+        return
+    pt.update()
+    it = pt.get("iteration", 0)
+    blob = pt.get("model")
+    if blob is not None and it != STATE["iteration"]:
+        STATE["model"] = deserialize_model(blob)
+        STATE["iteration"] = it
+
+associative_reducer = True
+commutative_reducer = True
+idempotent_reducer = True
+
+
+def _make_model():
+    g = torch.Generator().manual_seed(_CFG["seed"])
+    m = torch.nn.Sequential(
+        torch.nn.Linear(64, _CFG["hidden"]),
+        torch.nn.Tanh(),
+        torch.nn.Linear(_CFG["hidden"], 10),
+    )
+    with torch.no_grad():
+        for p in m.parameters():
+            p.copy_(torch.randn(p.shape, generator=g) * 0.1)
+    return m
+
+
+def _shard_batch(shard: int, iteration: int):
+    """Synthetic 8x8 'digits' batch, deterministic per (shard, iter)."""
+    g = torch.Generator().manual_seed(1000 * iteration + shard)
+    x = torch.randn(_CFG["bunch"], 64, generator=g)
+    y = torch.randint(0, 10, (_CFG["bunch"],), generator=g)
+    return x, y
+
+
+def serialize_model(m) -> str:
+    buf = io.BytesIO()
+    torch.save(m.state_dict(), buf)
+    return base64.b64encode(buf.getvalue()).decode()
+
+
+def deserialize_model(blob: str):
+    m = _make_model()
+    m.load_state_dict(torch.load(io.BytesIO(base64.b64decode(blob)),
+                                 weights_only=True))
+    return m
+
+
+def init(arg):
+    if arg:
+        _CFG.update({k: v for k, v in arg.items() if k in _CFG})
+    if STATE["model"] is None:
+        STATE["model"] = _make_model()
+        STATE["iteration"] = 0
+        STATE["losses"] = []
+
+
+def taskfn(emit):
+    for s in range(_CFG["shards"]):
+        emit(s + 1, {"shard": s})
+
+
+import threading
+
+_MAP_LOCK = threading.Lock()  # threaded local mode shares one model
+                              # replica; the reference's workers are
+                              # processes, so this matches its semantics
+
+
+def mapfn(key, value, emit):
+    """Gradient step on this shard (common.lua:85-104 analogue)."""
+    with _MAP_LOCK:
+        _sync_model_from_pt()
+        m = STATE["model"]
+        m.zero_grad()
+        x, y = _shard_batch(value["shard"], STATE["iteration"])
+        loss = torch.nn.functional.cross_entropy(m(x), y)
+        loss.backward()
+        grads = [(name, p.grad.detach().clone())
+                 for name, p in m.named_parameters()]
+    for name, g in grads:
+        emit(name, g)
+    emit("__loss__", torch.tensor([float(loss), 1.0]))
+
+
+def partitionfn(key):
+    from mapreduce_amd.utils.tuple import fnv1a32
+    return fnv1a32(key) % 4
+
+
+def reducefn(key, values, emit):
+    """Gradient sum (K6: axpy accumulation, common.lua:127-136)."""
+    acc = values[0].clone()
+    for v in values[1:]:
+        acc += v
+    emit(acc)
+
+
+combinerfn = reducefn
+
+
+def finalfn(pairs):
+    """Server side: SGD step on the summed gradients, then loop
+    (common.lua:144-202)."""
+    m = STATE["model"]
+    grads = {}
+    loss = None
+    for key, vals in pairs:
+        if key == "__loss__":
+            loss = vals[0]
+        else:
+            grads[key] = vals[0]
+    nshards = _CFG["shards"]
+    with torch.no_grad():
+        for name, p in m.named_parameters():
+            p -= _CFG["lr"] * grads[name] / nshards
+    STATE["losses"].append(float(loss[0] / loss[1]))
+    STATE["iteration"] += 1
+    pt = _pt()
+    if pt is not None:
+        pt.set("model", serialize_model(m))
+        pt.set("iteration", STATE["iteration"])
+        pt.update()
+    return "loop" if STATE["iteration"] < _CFG["iters"] else True

@@ -1,0 +1,204 @@
+"""Inverted index: word -> sorted list of (doc id, term frequency).
+
+The value-list-heavy workload of BASELINE.json (stresses the all-to-all
+shuffle with large value lists).  Map emits (word, doc); the "reduce"
+is a group-by — with the sort-once design the doc list of a key IS its
+contiguous segment, so reduce is segment bookkeeping, not computation
+(SURVEY.md K4 -> K1+K5).
+
+Pipeline per rank:
+  tokenize_spill -> (hash, pos); doc id = searchsorted(split offsets, pos)
+  composite sort: LSD radix by doc (low bits) then stable by hash
+  segment by (hash, doc) -> per-(word,doc) term frequency
+  [world > 1] exchange by hash partition (mulhi), re-sort, re-segment
+  segment by hash -> doc-list offsets per word + exemplar bytes
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import List, Optional, Tuple
+
+import torch
+
+from .. import ops
+from . import dist as dx
+
+
+@dataclass
+class InvertedIndexResult:
+    keys: torch.Tensor        # i64 u64-hash bits, sorted, one per word
+    doc_offsets: torch.Tensor  # i64[nwords+1] into docs/tf
+    docs: torch.Tensor        # i64: doc ids, sorted within each word
+    tf: torch.Tensor          # i64: term frequency of (word, doc)
+    pos: torch.Tensor         # exemplar (off<<16|len) into blob_src
+    blob_src: torch.Tensor    # u8
+
+    def to_host(self):
+        lens, blob = ops.extract_words(self.blob_src, self.pos)
+        raw = bytes(blob.cpu().numpy().tobytes())
+        offs = self.doc_offsets.cpu().tolist()
+        docs = self.docs.cpu().tolist()
+        tf = self.tf.cpu().tolist()
+        out = {}
+        off = 0
+        for i, L in enumerate(lens.cpu().tolist()):
+            w = raw[off:off + L]
+            off += L
+            out[w] = list(zip(docs[offs[i]:offs[i + 1]],
+                              tf[offs[i]:offs[i + 1]]))
+        return out
+
+
+class InvertedIndexJob:
+    def __init__(self, device, group=None, doc_base: int = 0):
+        self.device = torch.device(device)
+        self.group = group
+        self.rank, self.world = dx.world_info(group)
+        self.doc_base = doc_base  # global id of this rank's first doc
+
+    def _sort_by_doc_then_hash(self, h, d, p):
+        # LSD composite: stable sorts, least-significant key first
+        d1, h1, p1 = (d, h, p)
+        if h.is_cuda:
+            d1, h1, p1 = ops.sort_by_key(d, h, p, bits=32)
+            h2, d2, p2 = ops.sort_by_key(h1, d1, p1, bits=64)
+        else:
+            d1, h1, p1 = ops.sort_by_key(d, h, p, bits=64)
+            h2, d2, p2 = ops.sort_by_key(h1, d1, p1, bits=64)
+        return h2, d2, p2
+
+    def _segment_pairs(self, h, d, p):
+        """(hash, doc) grouped -> unique (hash, doc, tf, first pos)."""
+        # composite key for boundary detection: (hash, doc) change points.
+        # build u64 composite via xor-mix is unsafe; detect heads where
+        # hash or doc changes — reuse head_flags on each and OR.
+        if h.numel() == 0:
+            z = h
+            return h, d, torch.zeros_like(h), p
+        if h.is_cuda:
+            fh = ops.ext().head_flags(h)
+            fd = ops.ext().head_flags(d)
+            flags = (fh | fd)
+            seg = torch.cumsum(flags, 0)
+            nseg = int(seg[-1].item())
+            uh, tf = ops.ext().seg_reduce_i64(
+                h, torch.empty(0, dtype=torch.int64, device=h.device), seg,
+                nseg)
+            ud = ops.ext().seg_first_u64(d, seg, nseg)
+            up = ops.ext().seg_first_u64(p, seg, nseg)
+            return uh, ud, tf, up
+        import numpy as np
+        hn = h.numpy().view(np.uint64)
+        dn = d.numpy()
+        heads = np.ones(len(hn), dtype=bool)
+        heads[1:] = (hn[1:] != hn[:-1]) | (dn[1:] != dn[:-1])
+        idx = np.flatnonzero(heads)
+        tf = np.diff(np.append(idx, len(hn)))
+        from mapreduce_amd.ops._cpu import _from_u64
+        return (_from_u64(hn[idx]), torch.from_numpy(dn[idx]),
+                torch.from_numpy(tf.astype(np.int64)), p[torch.from_numpy(idx)])
+
+    def run(self, text: torch.Tensor,
+            splits: Optional[List[Tuple[int, int]]] = None
+            ) -> InvertedIndexResult:
+        dev = self.device
+        if splits is None:
+            splits = [(0, int(text.numel()))]
+        starts = torch.tensor([s for s, _ in splits] + [splits[-1][1]],
+                              device=dev, dtype=torch.int64)
+        if dev.type == "cuda":
+            cap = text.numel() // 2 + 16
+            h, p, c = ops.ext().tokenize_spill(text, 0, cap)
+            n = int(c.item())
+            h, p = h[:n], p[:n]
+        else:
+            h, p, n = ops.tokenize_words(text)
+        byte_off = p >> 16
+        d = torch.searchsorted(starts, byte_off, right=True) - 1
+        d = d + self.doc_base
+        h, d, p = self._sort_by_doc_then_hash(h, d, p)
+        uh, ud, tf, up = self._segment_pairs(h, d, p)
+
+        blob_src = text
+        if self.world > 1:
+            counts_d = ops.partition_counts(uh, self.world)
+            lens, blob = ops.extract_words(text, up)
+            bnd = torch.cumsum(counts_d, 0)
+            if lens.numel():
+                lcs = torch.cumsum(lens, 0)
+                cum = torch.where(
+                    bnd > 0, lcs.index_select(0, (bnd - 1).clamp(min=0)),
+                    torch.zeros_like(bnd))
+                blob_counts_d = torch.cat([cum[:1], cum[1:] - cum[:-1]])
+            else:
+                blob_counts_d = torch.zeros_like(counts_d)
+            recv_counts = dx.exchange_counts(counts_d, self.group)
+            recv_blob = dx.exchange_counts(blob_counts_d, self.group)
+            sc, rc = counts_d.cpu().tolist(), recv_counts.cpu().tolist()
+            sb, rb = blob_counts_d.cpu().tolist(), recv_blob.cpu().tolist()
+            rh = dx.exchange(uh, sc, rc, self.group)
+            rd = dx.exchange(ud, sc, rc, self.group)
+            rtf = dx.exchange(tf, sc, rc, self.group)
+            rlens = dx.exchange(lens, sc, rc, self.group)
+            rblob = dx.exchange(blob, sb, rb, self.group)
+            roff = torch.cumsum(rlens, 0) - rlens
+            rp = (roff << 16) | rlens
+            # composite re-sort (carrying pos + tf via an index payload) +
+            # merge duplicate (hash, doc) pairs arriving from several ranks
+            idx = torch.arange(rh.numel(), device=dev, dtype=torch.int64)
+            hs, ds, perm = self._sort_by_doc_then_hash(rh, rd, idx)
+            ps = rp.index_select(0, perm)
+            tfs = rtf.index_select(0, perm)
+            if hs.is_cuda and hs.numel():
+                fh = ops.ext().head_flags(hs)
+                fd = ops.ext().head_flags(ds)
+                seg = torch.cumsum(fh | fd, 0)
+                nseg = int(seg[-1].item())
+                uh, stf = ops.ext().seg_reduce_i64(hs, tfs, seg, nseg)
+                ud = ops.ext().seg_first_u64(ds, seg, nseg)
+                up = ops.ext().seg_first_u64(ps, seg, nseg)
+                tf = stf
+            else:
+                import numpy as np
+                hn = hs.numpy().view(np.uint64)
+                dn = ds.numpy()
+                heads = np.ones(len(hn), dtype=bool)
+                if len(hn):
+                    heads[1:] = (hn[1:] != hn[:-1]) | (dn[1:] != dn[:-1])
+                idx2 = np.flatnonzero(heads)
+                from mapreduce_amd.ops._cpu import _from_u64
+                seg_id = np.cumsum(heads) - 1
+                sums = np.zeros(len(idx2), dtype=np.int64)
+                np.add.at(sums, seg_id, tfs.numpy())
+                uh = _from_u64(hn[idx2])
+                ud = ds[torch.from_numpy(idx2)]
+                up = ps[torch.from_numpy(idx2)]
+                tf = torch.from_numpy(sums)
+            blob_src = rblob
+
+        # per-word doc-list offsets + one exemplar per word
+        seg, nwords = ops.segment_boundaries(uh)
+        if nwords:
+            if uh.is_cuda:
+                wk, doc_counts = ops.ext().seg_reduce_i64(
+                    uh, torch.empty(0, dtype=torch.int64, device=dev), seg,
+                    nwords)
+                wpos = ops.ext().seg_first_u64(up, seg, nwords)
+            else:
+                import numpy as np
+                hn = uh.numpy().view(np.uint64)
+                wkeys, idx3, cnts = np.unique(hn, return_index=True,
+                                              return_counts=True)
+                from mapreduce_amd.ops._cpu import _from_u64
+                wk = _from_u64(wkeys)
+                doc_counts = torch.from_numpy(cnts.astype(np.int64))
+                wpos = up[torch.from_numpy(idx3)]
+            offsets = torch.zeros(nwords + 1, dtype=torch.int64, device=dev)
+            torch.cumsum(doc_counts, 0, out=offsets[1:])
+        else:
+            wk = uh
+            wpos = up
+            offsets = torch.zeros(1, dtype=torch.int64, device=dev)
+        return InvertedIndexResult(keys=wk, doc_offsets=offsets, docs=ud,
+                                   tf=tf, pos=wpos, blob_src=blob_src)

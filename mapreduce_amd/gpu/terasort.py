@@ -1,0 +1,78 @@
+"""TeraSort: distributed sort of 64-bit keys (+64-bit payloads).
+
+The pure K1+K4 path of SURVEY.md: the reference's table.sort + heap-merge
+becomes one radix partition pass (top byte -> rank buckets, xGMI all-to-all)
+followed by a full local LSD radix sort.  Global order = rank-major (ranks
+own contiguous top-byte ranges) x locally sorted.
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+from .. import ops
+from . import dist as dx
+
+
+def _bucket_to_rank(world: int, device) -> torch.Tensor:
+    """bucket b (top byte) -> rank floor(b * world / 256); monotonic, so
+    rank key-ranges are contiguous."""
+    b = torch.arange(256, device=device, dtype=torch.int64)
+    return (b * world) >> 8
+
+
+class TeraSortJob:
+    def __init__(self, device, group=None):
+        self.device = torch.device(device)
+        self.group = group
+        self.rank, self.world = dx.world_info(group)
+
+    def run(self, keys: torch.Tensor,
+            payloads: Optional[torch.Tensor] = None
+            ) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
+        """keys: i64 (u64 bit order); returns this rank's globally-ordered
+        shard (rank-major partitioning over the sorted key space)."""
+        dev = self.device
+        if self.world > 1:
+            if dev.type == "cuda":
+                pl = payloads if payloads is not None else torch.empty(
+                    0, dtype=torch.int64, device=dev)
+                bk, bv, totals = ops.ext().radix_pass(keys, pl, 56)
+            else:
+                bk, bv = ops.sort_pairs(keys, payloads, bits=64)
+                # per-top-byte totals on CPU
+                import numpy as np
+                tb = (bk.numpy().view(np.uint64) >> np.uint64(56)).astype(
+                    np.int64)
+                totals = torch.from_numpy(
+                    np.bincount(tb, minlength=256).astype(np.int64))
+            # bucket -> rank send counts
+            b2r = _bucket_to_rank(self.world, totals.device)
+            send = torch.zeros(self.world, dtype=torch.int64,
+                               device=totals.device)
+            send.scatter_add_(0, b2r, totals)
+            recv = dx.exchange_counts(send.to(dev), self.group)
+            sc = send.cpu().tolist()
+            rc = recv.cpu().tolist()
+            keys = dx.exchange(bk, sc, rc, self.group)
+            if payloads is not None:
+                payloads = dx.exchange(bv, sc, rc, self.group)
+        sk, sv = ops.sort_pairs(keys, payloads, bits=64)
+        return sk, sv
+
+    def validate(self, sk: torch.Tensor) -> bool:
+        """Local sortedness in u64 bit order (cross-rank ordering follows
+        from the monotonic bucket->rank map)."""
+        if sk.numel() < 2:
+            return True
+        if sk.is_cuda:
+            flags = ops.ext().head_flags(sk)
+            # head_flags only says !=; check order via CPU on a sample +
+            # full check through sort idempotence
+            s2, _ = ops.sort_pairs(sk, None, bits=64)
+            return bool(torch.equal(s2, sk))
+        import numpy as np
+        a = sk.numpy().view(np.uint64)
+        return bool(np.all(a[:-1] <= a[1:]))

@@ -94,7 +94,7 @@ def sort_pairs(keys: torch.Tensor, vals: Optional[torch.Tensor], bits: int):
     ku = _u64(keys)
     order = np.argsort(ku, kind="stable")
     sk = _from_u64(ku[order])
-    if vals is None or vals.numel() == 0:
+    if vals is None:
         return sk, None
     return sk, vals[torch.from_numpy(order.astype(np.int64))]
 

@@ -27,6 +27,13 @@ from typing import Any, Callable, Iterable, Iterator, List, Tuple
 
 from .heap import Heap
 
+# the .tuple submodule shadows the builtin `tuple` in this namespace once
+# imported (e.g. via the package facade) — bind the builtins explicitly so
+# type checks below never silently compare against the module object
+_tuple = tuple
+_list = list
+_dict = dict
+
 # ---------------------------------------------------------------------------
 # Job status state machine (utils.lua:33-40)
 # ---------------------------------------------------------------------------
@@ -91,8 +98,10 @@ def sort_key(key: Any):
     by type so mixed-type key spaces have a stable global order.
     """
     t = type(key)
-    if t is tuple:
-        return (3, tuple(sort_key(k) for k in key))
+    if t is _tuple:
+        # length-first, then element-wise — matching InternedTuple's (and
+        # the reference tuple's) __lt ordering (tuple.lua:183-201)
+        return (3, len(key), _tuple(sort_key(k) for k in key))
     r = _TYPE_RANK.get(t)
     if r is None:
         from .tuple import InternedTuple
@@ -103,11 +112,8 @@ def sort_key(key: Any):
             r = 1
         elif isinstance(key, bytes):
             r = 2
-        elif isinstance(key, (tuple, InternedTuple)):
-            # interned composite keys order like plain tuples; the
-            # length-first ordering of InternedTuple itself only applies
-            # when comparing tuples directly (tuple.lua:183-201)
-            return (3, tuple(sort_key(k) for k in key))
+        elif isinstance(key, (_tuple, InternedTuple)):
+            return (3, len(key), _tuple(sort_key(k) for k in key))
         else:
             raise TypeError(f"unsupported key type: {t!r}")
     return (r, key)
@@ -123,11 +129,11 @@ def assert_check(value: Any, path: str = "value") -> None:
     the JSON-compatibility validation at utils.lua:313-333)."""
     if value is None or isinstance(value, (int, float, str, bytes, bool)):
         return
-    if isinstance(value, (list, tuple)):
+    if isinstance(value, (_list, _tuple)):
         for i, v in enumerate(value):
             assert_check(v, f"{path}[{i}]")
         return
-    if isinstance(value, dict):
+    if isinstance(value, _dict):
         for k, v in value.items():
             if not isinstance(k, (int, float, str, bytes, bool)):
                 raise TypeError(f"{path}: unsupported dict key {type(k)!r}")

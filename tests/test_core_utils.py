@@ -85,6 +85,18 @@ def test_sort_key_mixed_types():
     assert keys_sorted({k: 1 for k in keys}) == s
 
 
+def test_sort_key_nested_after_tuple_module_import():
+    # regression: importing mapreduce_amd.utils.tuple shadows the builtin
+    # `tuple` inside the utils package namespace; sort_key must still
+    # recurse into nested/mixed tuples
+    import mapreduce_amd.utils.tuple  # noqa: F401 (force the shadowing)
+    ks = [(2, "b"), (1, (2, 3)), (1, "a"), (10,)]
+    s = sorted(ks, key=sort_key)
+    assert s == [(10,), (1, "a"), (1, (2, 3)), (2, "b")]
+    it = tuple_(1, (2, 3))
+    assert sort_key(it) == sort_key((1, (2, 3)))
+
+
 def test_record_roundtrip():
     buf = io.BytesIO()
     rows = [("a", [1]), ((1, 2), [1, 2, 3]), (5, ["x"])]

@@ -1,0 +1,61 @@
+"""GPU numerics for TeraSort and inverted index (single rank)."""
+
+import collections
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def dev():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from mapreduce_amd import ops
+    ops.require_gpu_ext()
+    return torch.device("cuda:0")
+
+
+def test_terasort_gpu_large(dev):
+    from mapreduce_amd.gpu.terasort import TeraSortJob
+    rng = np.random.default_rng(31)
+    n = 4_000_000
+    keys = rng.integers(0, 2 ** 64 - 1, size=n, dtype=np.uint64)
+    pay = rng.integers(0, 2 ** 63, size=n, dtype=np.uint64)
+    job = TeraSortJob(dev)
+    sk, sv = job.run(torch.from_numpy(keys.view(np.int64)).to(dev),
+                     torch.from_numpy(pay.view(np.int64)).to(dev))
+    order = np.argsort(keys, kind="stable")
+    assert np.array_equal(sk.cpu().numpy().view(np.uint64), keys[order])
+    assert np.array_equal(sv.cpu().numpy().view(np.uint64), pay[order])
+
+
+def py_inverted_index(docs):
+    idx = {}
+    for d, text in enumerate(docs):
+        for w in text.split():
+            ent = idx.setdefault(w, {})
+            ent[d] = ent.get(d, 0) + 1
+    return {w: sorted(v.items()) for w, v in idx.items()}
+
+
+def test_inverted_index_gpu(dev):
+    from mapreduce_amd.gpu.inverted_index import InvertedIndexJob
+    rng = np.random.default_rng(33)
+    vocab = [f"term{i}q".encode() for i in range(500)]
+    docs = []
+    for _ in range(12):
+        ids = rng.integers(0, len(vocab), size=2000)
+        docs.append(b" ".join(vocab[i] for i in ids.tolist()))
+    blob = b" ".join(docs) + b" "
+    offs = [0]
+    for d in docs[:-1]:
+        offs.append(offs[-1] + len(d) + 1)
+    offs.append(len(blob))
+    splits = list(zip(offs[:-1], offs[1:]))
+    text = torch.from_numpy(np.frombuffer(blob, dtype=np.uint8).copy()).to(dev)
+    job = InvertedIndexJob(dev)
+    res = job.run(text, splits)
+    assert res.to_host() == py_inverted_index(docs)

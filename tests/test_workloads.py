@@ -1,0 +1,146 @@
+"""TeraSort + inverted index on the CPU tier (engine logic; GPU numerics
+covered by test_gpu_workloads.py), single-rank and gloo world_size=2."""
+
+import collections
+import os
+import socket
+
+import numpy as np
+import pytest
+import torch
+
+from mapreduce_amd.gpu.inverted_index import InvertedIndexJob
+from mapreduce_amd.gpu.terasort import TeraSortJob
+
+
+def _u64(t):
+    return t.numpy().view(np.uint64)
+
+
+def test_terasort_single_rank_cpu():
+    rng = np.random.default_rng(5)
+    keys = rng.integers(0, 2 ** 64 - 1, size=50_000, dtype=np.uint64)
+    pay = np.arange(50_000, dtype=np.uint64)
+    job = TeraSortJob("cpu")
+    sk, sv = job.run(torch.from_numpy(keys.view(np.int64)),
+                     torch.from_numpy(pay.view(np.int64)))
+    order = np.argsort(keys, kind="stable")
+    assert np.array_equal(_u64(sk), keys[order])
+    assert np.array_equal(_u64(sv), pay[order])
+    assert job.validate(sk)
+
+
+def py_inverted_index(docs):
+    idx = {}
+    for d, text in enumerate(docs):
+        for w in text.split():
+            ent = idx.setdefault(w, {})
+            ent[d] = ent.get(d, 0) + 1
+    return {w: sorted(v.items()) for w, v in idx.items()}
+
+
+def test_inverted_index_single_rank_cpu():
+    rng = np.random.default_rng(9)
+    vocab = [f"tok{i}".encode() for i in range(200)]
+    docs = []
+    for _ in range(6):
+        ids = rng.integers(0, len(vocab), size=500)
+        docs.append(b" ".join(vocab[i] for i in ids.tolist()))
+    blob = b" ".join(docs) + b" "
+    # split offsets: each doc starts right after the previous separator
+    offs = [0]
+    for d in docs[:-1]:
+        offs.append(offs[-1] + len(d) + 1)
+    offs.append(len(blob))
+    splits = list(zip(offs[:-1], offs[1:]))
+    text = torch.from_numpy(np.frombuffer(blob, dtype=np.uint8).copy())
+    job = InvertedIndexJob("cpu")
+    res = job.run(text, splits)
+    got = res.to_host()
+    exp = py_inverted_index(docs)
+    assert got == exp
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _ts_worker(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.distributed.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        rng = np.random.default_rng(40 + rank)
+        keys = rng.integers(0, 2 ** 64 - 1, size=20_000, dtype=np.uint64)
+        job = TeraSortJob("cpu")
+        sk, _ = job.run(torch.from_numpy(keys.view(np.int64)), None)
+        assert job.validate(sk)
+        mine = _u64(sk)
+        # ownership: my keys' top bytes map to my rank
+        if len(mine):
+            tb = (mine >> np.uint64(56)).astype(np.int64)
+            assert ((tb * world) >> 8 == rank).all()
+        # global preservation
+        all_keys = [None] * world
+        torch.distributed.all_gather_object(all_keys, mine.tolist())
+        all_in = [None] * world
+        torch.distributed.all_gather_object(all_in, keys.tolist())
+        if rank == 0:
+            got = sorted(x for l in all_keys for x in l)
+            exp = sorted(x for l in all_in for x in l)
+            assert got == exp
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_terasort_gloo_ws2():
+    torch.multiprocessing.spawn(_ts_worker, args=(2, _free_port()),
+                                nprocs=2, join=True)
+
+
+def _ii_worker(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.distributed.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        rng = np.random.default_rng(70 + rank)
+        vocab = [f"w{i}".encode() for i in range(60)]
+        docs = []
+        for _ in range(3):
+            ids = rng.integers(0, len(vocab), size=200)
+            docs.append(b" ".join(vocab[i] for i in ids.tolist()))
+        blob = b" ".join(docs) + b" "
+        offs = [0]
+        for d in docs[:-1]:
+            offs.append(offs[-1] + len(d) + 1)
+        offs.append(len(blob))
+        splits = list(zip(offs[:-1], offs[1:]))
+        text = torch.from_numpy(np.frombuffer(blob, dtype=np.uint8).copy())
+        job = InvertedIndexJob("cpu", doc_base=rank * 3)
+        res = job.run(text, splits)
+        part = res.to_host()
+        alldocs = [None] * world
+        torch.distributed.all_gather_object(alldocs, docs)
+        allparts = [None] * world
+        torch.distributed.all_gather_object(allparts, part)
+        if rank == 0:
+            merged = {}
+            for p in allparts:
+                for w, lst in p.items():
+                    assert w not in merged, "word owned by two ranks"
+                    merged[w] = lst
+            flat = [d for dl in alldocs for d in dl]
+            assert merged == py_inverted_index(flat)
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_inverted_index_gloo_ws2():
+    torch.multiprocessing.spawn(_ii_worker, args=(2, _free_port()),
+                                nprocs=2, join=True)
