@@ -62,6 +62,7 @@ def main() -> int:
 
     def one_step():
         res = job.run(corpus.text, splits)
+        res.materialize()  # results land in host memory every step (C7/C8)
         return res
 
     # warmup (untimed)

@@ -40,6 +40,14 @@ class WordCountResult:
     blob_src: torch.Tensor  # u8 source for exemplar bytes
     nwords: int             # words processed by this rank this step
 
+    def materialize(self):
+        """Deliver the job's results to host memory (the analogue of the
+        reference writing result.P<p> files + the server reading them,
+        C7/C8): raw key/count arrays + the packed exemplar word bytes.
+        Returns (keys_cpu, counts_cpu, lens_cpu, blob_cpu)."""
+        lens, blob = ops.extract_words(self.blob_src, self.pos)
+        return (self.keys.cpu(), self.counts.cpu(), lens.cpu(), blob.cpu())
+
     def to_host(self) -> List[Tuple[bytes, int]]:
         """Materialize (word, count) pairs, sorted by key hash (C8)."""
         lens, blob = ops.extract_words(self.blob_src, self.pos)
