@@ -39,6 +39,8 @@ def main() -> int:
     p.add_argument("--device", default=None)
     p.add_argument("--mode", default="auto",
                    choices=["auto", "streaming", "fused"])
+    p.add_argument("--uncoordinated", action="store_true",
+                   help="bypass the control plane (engine-only timing)")
     args = p.parse_args()
 
     from mapreduce_amd import ops
@@ -55,13 +57,23 @@ def main() -> int:
     job = WordCountJob(device, vocab_estimate=max(args.vocab, 1 << 12),
                        mode=args.mode)
     splits = corpus.splits()
+    runner = None
+    if not args.uncoordinated:
+        from mapreduce_amd.gpu.runner import GpuClusterRunner
+
+        runner = GpuClusterRunner(job, claim_mode="batch")
 
     def sync():
         if device.type == "cuda":
             torch.cuda.synchronize(device)
 
     def one_step():
-        res = job.run(corpus.text, splits)
+        # full-framework step: map jobs tracked + claimed through the
+        # control plane, engine executes, collective shuffle+reduce
+        if runner is not None:
+            res = runner.run(corpus.text, splits)
+        else:
+            res = job.run(corpus.text, splits)
         res.materialize()  # results land in host memory every step (C7/C8)
         return res
 

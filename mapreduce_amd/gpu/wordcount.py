@@ -86,31 +86,53 @@ class WordCountJob:
                                     self.device)
         self._nwords.zero_()
 
-    def _map_combine_streaming(self, text: torch.Tensor, s0: int, e0: int
-                               ) -> int:
-        """Streaming map+combine: tokenize with per-block LDS caches (Zipf
-        head aggregates in LDS); cache misses spill as (hash,pos), get
-        bucketized by top byte (one radix pass) and LDS-counted per bucket.
-        No per-word global-memory probe on any path.  Returns the word
-        count (the one host sync of the map phase)."""
-        chunk = text[s0:e0]
-        cap = chunk.numel() // 2 + 16
-        h, p, c = ops.ext().tokenize_cache_spill(
-            chunk, s0, self.table.tkeys, self.table.tvals, self.table.texm,
-            cap, self._nwords)
-        cnt = c.item()          # spill count (syncs; nwords rides along)
-        n = int(self._nwords.item())
-        nspill = int(cnt)
-        if nspill:
-            h, p = h[:nspill], p[:nspill]
-            hk, pv, totals = ops.ext().radix_pass(h, p, 56)
-            bucket_off = torch.zeros(257, dtype=torch.int64,
-                                     device=self.device)
-            torch.cumsum(totals, 0, out=bucket_off[1:])
-            ops.ext().bucket_count(hk, pv, bucket_off, 256, 8,
-                                   self.table.tkeys, self.table.tvals,
-                                   self.table.texm)
-        return n
+    # ---------------- claimable phase API (used by gpu.runner) -----------
+    # begin_map() -> map_split(s, e) per claimed job -> finish_map() ->
+    # shuffle_reduce() — run() composes these; the cluster runner drives
+    # them under control-plane job claims.
+
+    def begin_map(self, text: torch.Tensor) -> None:
+        self.reset()
+        self._text = text
+        if self.mode == "streaming":
+            cap = text.numel() // 2 + 16
+            opts = dict(dtype=torch.int64, device=self.device)
+            self._spill_h = torch.empty(cap, **opts)
+            self._spill_p = torch.empty(cap, **opts)
+            self._spill_c = torch.zeros(1, **opts)
+            self._spill_cap = cap
+
+    def map_split(self, s: int, e: int) -> None:
+        """One map job: tokenize+combine bytes [s, e) of the corpus.
+        Streaming mode appends cache misses to the shared spill arrays
+        (atomic counter append composes across launches)."""
+        if self.mode == "streaming":
+            ops.ext().tokenize_cache_spill(
+                self._text[s:e], s, self.table.tkeys, self.table.tvals,
+                self.table.texm, self._spill_cap, self._nwords,
+                self._spill_h, self._spill_p, self._spill_c)
+        else:
+            self.table.tokenize_count(self._text[s:e], s, self._nwords)
+
+    def finish_map(self) -> int:
+        """Drain the spill through bucketize + per-bucket LDS count.
+        Returns the word count (the one host sync of the map phase)."""
+        if self.mode == "streaming":
+            nspill = int(self._spill_c.item())
+            n = int(self._nwords.item())
+            if nspill:
+                h = self._spill_h[:nspill]
+                p = self._spill_p[:nspill]
+                hk, pv, totals = ops.ext().radix_pass(h, p, 56)
+                bucket_off = torch.zeros(257, dtype=torch.int64,
+                                         device=self.device)
+                torch.cumsum(totals, 0, out=bucket_off[1:])
+                ops.ext().bucket_count(hk, pv, bucket_off, 256, 8,
+                                       self.table.tkeys, self.table.tvals,
+                                       self.table.texm)
+            self._spill_h = self._spill_p = None
+            return n
+        return int(self._nwords.item())
 
     @staticmethod
     def _coalesced(text: torch.Tensor, splits) -> bool:
@@ -126,27 +148,28 @@ class WordCountJob:
     def run(self, text: torch.Tensor,
             splits: Optional[List[Tuple[int, int]]] = None) -> WordCountResult:
         """One full job over this rank's corpus bytes."""
-        self.reset()
-        dev = self.device
         if splits is None:
             splits = [(0, int(text.numel()))]
-
         # ---- MAP + COMBINE.  Map jobs (splits) sharing one contiguous
         # corpus buffer are coalesced into a single fused kernel launch:
         # a launch needs >>256 workgroups to fill the chip, and 197 small
         # launches serialize (measured 98% of step time before fusing).
         # Split boundaries are whitespace-aligned, so tokenization over the
         # coalesced range is byte-identical to per-split runs.
-        nwords_host = None
-        if self.mode == "streaming" and self._coalesced(text, splits):
-            s0, e0 = splits[0][0], splits[-1][1]
-            nwords_host = self._map_combine_streaming(text, s0, e0)
-        elif self._coalesced(text, splits):
-            s0, e0 = splits[0][0], splits[-1][1]
-            self.table.tokenize_count(text[s0:e0], s0, self._nwords)
+        self.begin_map(text)
+        if self._coalesced(text, splits):
+            self.map_split(splits[0][0], splits[-1][1])
         else:
             for (s, e) in splits:
-                self.table.tokenize_count(text[s:e], s, self._nwords)
+                self.map_split(s, e)
+        nwords = self.finish_map()
+        return self.shuffle_reduce(nwords)
+
+    def shuffle_reduce(self, nwords: int) -> WordCountResult:
+        """Phase 2: extract + sort uniques, all-to-all exchange, segmented
+        reduce — collective across ranks (every rank must enter)."""
+        text = self._text
+        dev = self.device
 
         # ---- EXTRACT + SORT
         uk, uv, up = self.table.extract()
@@ -188,7 +211,5 @@ class WordCountJob:
             fk, fv, fp = sk, sv, sp
             blob_src = text
 
-        nwords = (nwords_host if nwords_host is not None
-                  else int(self._nwords.item()))
         return WordCountResult(keys=fk, counts=fv, pos=fp,
                                blob_src=blob_src, nwords=nwords)

@@ -94,19 +94,20 @@ std::vector<torch::Tensor> tokenize_spill(torch::Tensor text, long pos_base,
 }
 
 // -------------------------------------------------------------- K2+K5 v4
-std::vector<torch::Tensor> tokenize_cache_spill(
+// out arrays + counter are caller-owned so several map-job launches can
+// append into ONE spill stream (atomic counter composes across launches)
+void tokenize_cache_spill(
     torch::Tensor text, long pos_base, torch::Tensor tkeys,
     torch::Tensor tvals, torch::Tensor texm, long spill_cap,
-    torch::Tensor nwords) {
+    torch::Tensor nwords, torch::Tensor out_hash, torch::Tensor out_pos,
+    torch::Tensor counter) {
   TORCH_CHECK(text.is_cuda() && text.scalar_type() == torch::kUInt8 &&
               text.is_contiguous(), "text must be contiguous u8 on GPU");
   long n = text.numel();
   long cap = tkeys.numel();
   TORCH_CHECK((cap & (cap - 1)) == 0, "table capacity must be a power of 2");
-  auto opts = torch::TensorOptions().device(text.device()).dtype(torch::kInt64);
-  auto out_hash = torch::empty({spill_cap}, opts);
-  auto out_pos = torch::empty({spill_cap}, opts);
-  auto counter = torch::zeros({1}, opts);
+  TORCH_CHECK(out_hash.numel() >= spill_cap && out_pos.numel() >= spill_cap,
+              "spill arrays too small");
   if (n) {
     // default v4 (cache + spill in the scan loop).  The v5 word-list
     // restructure measured 2x SLOWER end-to-end (11.5 vs 6.0 ms/step,
@@ -135,7 +136,6 @@ std::vector<torch::Tensor> tokenize_cache_spill(
                          spill_cap,
                          reinterpret_cast<unsigned long long*>(nwords.data_ptr<i64>()));
   }
-  return {out_hash, out_pos, counter};
 }
 
 // diagnosis only
