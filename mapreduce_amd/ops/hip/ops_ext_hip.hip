@@ -10,6 +10,7 @@
 #include <ATen/hip/HIPContext.h>
 #include <torch/extension.h>
 
+#include "gradsum.hip"
 #include "mr_kernels.hip"
 #include "radix_sort.hip"
 
@@ -324,6 +325,27 @@ torch::Tensor gather_bytes(torch::Tensor text, torch::Tensor pos,
   return out;
 }
 
+// ------------------------------------------------------------------------ K6
+torch::Tensor grad_colsum(torch::Tensor grads, bool use_mfma) {
+  TORCH_CHECK(grads.is_cuda() && grads.dim() == 2 &&
+              grads.scalar_type() == torch::kFloat32 && grads.is_contiguous(),
+              "grads must be contiguous f32 [G, D] on GPU");
+  long G = grads.size(0), D = grads.size(1);
+  auto out = torch::empty({D}, grads.options());
+  if (use_mfma) {
+    long ntiles = (D + 15) / 16;
+    long blocks = ntiles < 4096 ? (ntiles ? ntiles : 1) : 4096;
+    hipLaunchKernelGGL(colsum_f32_mfma_kernel, dim3(blocks), dim3(64), 0,
+                       cur_stream(), grads.data_ptr<float>(), G, D,
+                       out.data_ptr<float>());
+  } else {
+    hipLaunchKernelGGL(colsum_f32_valu_kernel, dim3(grid_for(D, 4)),
+                       dim3(kBlock), 0, cur_stream(),
+                       grads.data_ptr<float>(), G, D, out.data_ptr<float>());
+  }
+  return out;
+}
+
 // ------------------------------------------------------------------------ K1
 // single partition pass on one 8-bit digit; returns (keys, vals, per-digit
 // totals) — used to bucketize by top byte before LDS counting
@@ -411,4 +433,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gather_bytes", &gather_bytes);
   m.def("radix_sort_pairs", &radix_sort_pairs);
   m.def("radix_pass", &radix_pass);
+  m.def("grad_colsum", &grad_colsum, "sum G gradient rows -> D (K6)");
 }
