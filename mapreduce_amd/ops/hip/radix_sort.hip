@@ -41,6 +41,104 @@ __global__ __launch_bounds__(RS_BLOCK) void radix_hist_kernel(
     hist[(long)b * ntiles + tile] = (i64)lh[b];
 }
 
+// v2: LDS-staged scatter.  v1 writes each element straight to its global
+// position — 16 B granules scattered across 256 digit destinations, so
+// nearly every write wastes most of its cache line.  v2 first reorders the
+// whole tile in LDS by (digit, stable rank), then streams it out in stage
+// order: consecutive lanes write consecutive positions of each digit run
+// (avg run = TILE/256 elements), restoring write coalescing.
+// digit_start_in_tile comes from a 256-entry LDS prefix over this tile's
+// histogram (recomputed; must equal radix_hist_kernel's counts).
+__global__ __launch_bounds__(RS_BLOCK) void radix_scatter_v2_kernel(
+    const u64* __restrict__ keys, const u64* __restrict__ vals, long n,
+    int shift, long ntiles, const i64* __restrict__ base_dx,
+    u64* __restrict__ okeys, u64* __restrict__ ovals) {
+  __shared__ u32 wavecnt[RS_WAVES][RS_BINS];
+  __shared__ u32 cnt_base[RS_BINS];     // running per-digit counts
+  __shared__ u32 digit_start[RS_BINS + 1];
+  __shared__ u64 stage_k[RS_TILE];
+  __shared__ u64 stage_v[RS_TILE];
+  for (int b = threadIdx.x; b < RS_BINS; b += blockDim.x) cnt_base[b] = 0;
+  long tile = blockIdx.x;
+  long tbase = tile * RS_TILE;
+  long tile_n = n - tbase < RS_TILE ? n - tbase : RS_TILE;
+  int wave = threadIdx.x / WAVE;
+  int lane = threadIdx.x % WAVE;
+  u64 lt_mask = ((u64)1 << lane) - 1;
+  bool has_vals = ovals != nullptr;
+
+  // ---- per-tile digit histogram + exclusive prefix -> digit_start
+  __syncthreads();
+  for (int r = 0; r < RS_ITEMS; ++r) {
+    long i = tbase + (long)r * RS_BLOCK + threadIdx.x;
+    if (i < n)
+      atomicAdd(&cnt_base[(u32)((keys[i] >> shift) & 0xFF)], 1u);
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    u32 run = 0;
+    for (int d = 0; d < RS_BINS; ++d) {  // serial 256-step scan: ~cheap
+      digit_start[d] = run;
+      run += cnt_base[d];
+    }
+    digit_start[RS_BINS] = run;
+  }
+  __syncthreads();
+  for (int b = threadIdx.x; b < RS_BINS; b += blockDim.x) cnt_base[b] = 0;
+  __syncthreads();
+
+  // ---- stable rank + stage into LDS at (digit_start + tile_rank)
+  for (int r = 0; r < RS_ITEMS; ++r) {
+    for (int b = threadIdx.x; b < RS_BINS; b += blockDim.x)
+      for (int w = 0; w < RS_WAVES; ++w) wavecnt[w][b] = 0;
+    __syncthreads();
+    long i = tbase + (long)r * RS_BLOCK + threadIdx.x;
+    bool valid = i < n;
+    u64 k = valid ? keys[i] : 0;
+    u32 d = valid ? (u32)((k >> shift) & 0xFF) : 0;
+    u64 m = __ballot(valid);
+    #pragma unroll
+    for (int b = 0; b < 8; ++b) {
+      u64 bb = __ballot(valid && ((d >> b) & 1));
+      m &= ((d >> b) & 1) ? bb : ~bb;
+    }
+    u32 lane_rank = 0;
+    if (valid) {
+      lane_rank = (u32)__popcll(m & lt_mask);
+      int leader = __ffsll((unsigned long long)m) - 1;
+      if (lane == leader) wavecnt[wave][d] = (u32)__popcll(m);
+    }
+    __syncthreads();
+    if (threadIdx.x < RS_BINS) {
+      u32 run = cnt_base[threadIdx.x];
+      #pragma unroll
+      for (int w = 0; w < RS_WAVES; ++w) {
+        u32 c = wavecnt[w][threadIdx.x];
+        wavecnt[w][threadIdx.x] = run;
+        run += c;
+      }
+      cnt_base[threadIdx.x] = run;
+    }
+    __syncthreads();
+    if (valid) {
+      u32 s = digit_start[d] + wavecnt[wave][d] + lane_rank;
+      stage_k[s] = k;
+      if (has_vals) stage_v[s] = vals[i];
+    }
+    __syncthreads();
+  }
+
+  // ---- stream out in stage order: coalesced within each digit run
+  for (long s = threadIdx.x; s < tile_n; s += blockDim.x) {
+    u64 k = stage_k[s];
+    u32 d = (u32)((k >> shift) & 0xFF);
+    u32 local = (u32)s - digit_start[d];
+    long pos = base_dx[(long)d * ntiles + tile] + local;
+    okeys[pos] = k;
+    if (has_vals) ovals[pos] = stage_v[s];
+  }
+}
+
 __global__ __launch_bounds__(RS_BLOCK) void radix_scatter_kernel(
     const u64* __restrict__ keys, const u64* __restrict__ vals, long n,
     int shift, long ntiles, const i64* __restrict__ base_dx,
