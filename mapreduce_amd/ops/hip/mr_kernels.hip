@@ -630,6 +630,153 @@ __global__ __launch_bounds__(256) void tokenize_v5_kernel(
   if (lane == 0 && ws) atomicAdd(nwords, ws);
 }
 
+// ---------------------------------------------------------------------------
+// K2+K5 v6: v4's single-phase structure + v5's branchless scan.
+// Per thread: classify its 16 LDS bytes (+16 lookahead) into a whitespace
+// bitmask with compile-time-unrolled extracts, iterate word starts with
+// ffs (~words, not ~bytes, iterations), hash from REGISTERS via funnel
+// shifts (zero LDS re-reads), insert into the v4 LDS cache, spill misses.
+// Same two barriers per tile as v4 — the v5 word-list experiment showed
+// extra barriers/LDS cost more than lane balance buys.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void tokenize_v6_kernel(
+    const u8* __restrict__ text, long n, u64 pos_base,
+    u64* __restrict__ tkeys, i64* __restrict__ tvals, u64* __restrict__ texm,
+    u64 cap_mask, u64* __restrict__ out_hash, u64* __restrict__ out_pos,
+    unsigned long long* __restrict__ spill_counter, long spill_cap,
+    unsigned long long* __restrict__ nwords) {
+  __shared__ __align__(16) u8 tile[TOK_TILE + TOK_HALO];
+  __shared__ u64 ckeys[TOK_CACHE];
+  __shared__ u64 cpos[TOK_CACHE];
+  __shared__ u32 ccnt[TOK_CACHE];
+  for (int s = threadIdx.x; s < TOK_CACHE; s += blockDim.x) {
+    ckeys[s] = HT_EMPTY;
+    ccnt[s] = 0;
+  }
+  const int tid = threadIdx.x;
+  unsigned long long my_words = 0;
+  long tile0 = (long)blockIdx.x * TOK_TILE;
+  long tstride = (long)gridDim.x * TOK_TILE;
+  for (long base = tile0; base < n; base += tstride) {
+    __syncthreads();
+    long avail = n - base;
+    long want = avail < TOK_TILE + TOK_HALO ? avail : TOK_TILE + TOK_HALO;
+    for (int o = tid * 16; o < want; o += blockDim.x * 16) {
+      if (o + 16 <= want && (((uintptr_t)&text[base + o]) & 15) == 0) {
+        *(uint4*)&tile[o] = *(const uint4*)&text[base + o];
+      } else {
+        for (int b = 0; b < 16 && o + b < want; ++b)
+          tile[o + b] = text[base + o + b];
+      }
+    }
+    __syncthreads();
+    long my0 = (long)tid * TOK_BYTES;
+    if (my0 >= avail) continue;
+    // 32 bytes in registers: my window + 16B lookahead (halo-staged)
+    const uint4 va = *(const uint4*)&tile[my0];
+    const uint4 vb = *(const uint4*)&tile[my0 + 16];
+    const u32 rs[8] = {va.x, va.y, va.z, va.w, vb.x, vb.y, vb.z, vb.w};
+    u32 m32 = 0;
+    #pragma unroll
+    for (int r = 0; r < 8; ++r)
+      #pragma unroll
+      for (int b = 0; b < 4; ++b)
+        m32 |= (u32)is_ws((u8)(rs[r] >> (8 * b))) << (r * 4 + b);
+    // bytes at/after `want` count as whitespace (stale stage data)
+    long lim32 = want - my0;
+    if (lim32 < 32)
+      m32 |= (lim32 <= 0) ? 0xFFFFFFFFu : ~((1u << lim32) - 1);
+    u32 prevb = (base + my0 == 0)
+                    ? 1u
+                    : (u32)is_ws(my0 ? tile[my0 - 1] : text[base - 1]);
+    u32 sm = ~m32 & ((m32 << 1) | prevb) & 0xFFFFu;
+    long limw = avail - my0;  // starts must be real data bytes
+    if (limw < 16) sm &= (limw <= 0) ? 0u : ((1u << limw) - 1);
+    my_words += __popc(sm);
+    const u64 q0 = (u64)va.x | ((u64)va.y << 32);
+    const u64 q1 = (u64)va.z | ((u64)va.w << 32);
+    const u64 q2 = (u64)vb.x | ((u64)vb.y << 32);
+    const u64 q3 = (u64)vb.z | ((u64)vb.w << 32);
+    u64 sh_[8];
+    u64 sp_[8];
+    int ns = 0;
+    while (sm) {
+      int s = __ffs(sm) - 1;
+      sm &= sm - 1;
+      u32 t = m32 >> s;
+      u64 h = FNV64_OFFSET;
+      u32 len;
+      if (t) {
+        len = (u32)(__ffs(t) - 1);  // bit s is clear, so len >= 1
+        // hash bytes s..s+len-1 from q0..q3 (s+len <= 32 by construction):
+        // branchless 4-way register select per byte — no LDS re-reads
+        for (u32 b2 = 0; b2 < len; ++b2) {
+          int j = s + (int)b2;
+          u64 lo = (j & 16) ? q2 : q0;
+          u64 hi = (j & 16) ? q3 : q1;
+          u64 w = (j & 8) ? hi : lo;
+          h = (h ^ ((w >> (8 * (j & 7))) & 0xFF)) * FNV64_PRIME;
+        }
+      } else {
+        // word longer than the 32-byte window: scan from global
+        long g = base + my0 + s;
+        while (g < n && !is_ws(text[g])) {
+          h = (h ^ text[g]) * FNV64_PRIME;
+          ++g;
+        }
+        len = (u32)((g - (base + my0 + s)) > 0xFFFF
+                        ? 0xFFFF : g - (base + my0 + s));
+      }
+      u64 k = remap_key(h);
+      u64 p = ((pos_base + (u64)(base + my0 + s)) << 16) | (u64)len;
+      // LDS cache insert (v4 path)
+      u32 slot = (u32)((k ^ (k >> 32)) & (TOK_CACHE - 1));
+      bool done = false;
+      for (int pr = 0; pr < TOK_PROBE; ++pr) {
+        u64 cur2 = ckeys[slot];
+        if (cur2 == k) {
+          atomicAdd(&ccnt[slot], 1u);
+          done = true;
+          break;
+        }
+        if (cur2 == HT_EMPTY) {
+          u64 prevk = atomicCAS((unsigned long long*)&ckeys[slot],
+                                (unsigned long long)HT_EMPTY,
+                                (unsigned long long)k);
+          if (prevk == HT_EMPTY) cpos[slot] = p;
+          if (prevk == HT_EMPTY || prevk == k) {
+            atomicAdd(&ccnt[slot], 1u);
+            done = true;
+            break;
+          }
+        }
+        slot = (slot + 1) & (TOK_CACHE - 1);
+      }
+      if (!done) {
+        sh_[ns] = k;
+        sp_[ns] = p;
+        ++ns;
+      }
+    }
+    if (ns) {
+      unsigned long long o = atomicAdd(spill_counter, (unsigned long long)ns);
+      for (int w = 0; w < ns; ++w)
+        if ((long)o + w < spill_cap) {
+          out_hash[o + w] = sh_[w];
+          out_pos[o + w] = sp_[w];
+        }
+    }
+  }
+  __syncthreads();
+  for (int s = tid; s < TOK_CACHE; s += blockDim.x)
+    if (ckeys[s] != HT_EMPTY && ccnt[s])
+      ht_add(ckeys[s], cpos[s], (i64)ccnt[s], tkeys, tvals, texm, cap_mask);
+  unsigned long long ws = my_words;
+  for (int off = 32; off > 0; off >>= 1) ws += __shfl_down(ws, off, WAVE);
+  if (lane_id() == 0 && ws) atomicAdd(nwords, ws);
+}
+
 // Ablation copy of the tokenize kernel (diagnosis only — §5.4 rule 8:
 // ablate empirically before optimizing).  mode: 1=stage tiles only,
 // 2=+word-boundary scan (count only), 3=+FNV hash, 4=+LDS cache insert,

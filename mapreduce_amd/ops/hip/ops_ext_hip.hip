@@ -137,7 +137,18 @@ void tokenize_cache_spill(
     // blocks/CU) and per-round ballots outweigh the balance win at this
     // grain.  Kept behind MR_TOKENIZE_V5=1 as a recorded experiment.
     const char* v = getenv("MR_TOKENIZE_V5");
-    if (!(v && v[0] == '1'))
+    const char* v6 = getenv("MR_TOKENIZE_V6");
+    if (v6 && v6[0] == '1')
+      hipLaunchKernelGGL(tokenize_v6_kernel,
+                         dim3(grid_for(n, TOK_BYTES)), dim3(kBlock), 0,
+                         cur_stream(), text.data_ptr<u8>(), n, (u64)pos_base,
+                         u64p(tkeys), tvals.data_ptr<i64>(),
+                         texm.numel() ? u64p(texm) : nullptr, (u64)(cap - 1),
+                         u64p(out_hash), u64p(out_pos),
+                         reinterpret_cast<unsigned long long*>(counter.data_ptr<i64>()),
+                         spill_cap,
+                         reinterpret_cast<unsigned long long*>(nwords.data_ptr<i64>()));
+    else if (!(v && v[0] == '1'))
       hipLaunchKernelGGL(tokenize_cache_spill_kernel,
                          dim3(grid_for(n, TOK_BYTES)), dim3(kBlock), 0,
                          cur_stream(), text.data_ptr<u8>(), n, (u64)pos_base,
