@@ -131,14 +131,15 @@ void tokenize_cache_spill(
   TORCH_CHECK(out_hash.numel() >= spill_cap && out_pos.numel() >= spill_cap,
               "spill arrays too small");
   if (n) {
-    // default v4 (cache + spill in the scan loop).  The v5 word-list
-    // restructure measured 2x SLOWER end-to-end (11.5 vs 6.0 ms/step,
-    // same box): its extra barriers, LDS footprint (+12 KB -> fewer
-    // blocks/CU) and per-round ballots outweigh the balance win at this
-    // grain.  Kept behind MR_TOKENIZE_V5=1 as a recorded experiment.
+    // Default v6: branchless ws-mask scan + register-funnel hashing in
+    // v4's single-phase structure — measured 5.10 vs 6.25 ms/step (+23%)
+    // over v4's per-byte scan.  Recorded alternatives: MR_TOKENIZE_V4=1
+    // (per-byte scan loop), MR_TOKENIZE_V5=1 (word-list restructure,
+    // measured 2x SLOWER: extra barriers + LDS footprint outweigh lane
+    // balance at this grain).
     const char* v = getenv("MR_TOKENIZE_V5");
-    const char* v6 = getenv("MR_TOKENIZE_V6");
-    if (v6 && v6[0] == '1')
+    const char* v4 = getenv("MR_TOKENIZE_V4");
+    if (!(v && v[0] == '1') && !(v4 && v4[0] == '1'))
       hipLaunchKernelGGL(tokenize_v6_kernel,
                          dim3(grid_for(n, TOK_BYTES)), dim3(kBlock), 0,
                          cur_stream(), text.data_ptr<u8>(), n, (u64)pos_base,
