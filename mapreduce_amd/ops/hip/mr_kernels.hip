@@ -640,6 +640,7 @@ __global__ __launch_bounds__(256) void tokenize_v5_kernel(
 // extra barriers/LDS cost more than lane balance buys.
 // ---------------------------------------------------------------------------
 
+template <int TOK_CACHE>
 __global__ __launch_bounds__(256) void tokenize_v6_kernel(
     const u8* __restrict__ text, long n, u64 pos_base,
     u64* __restrict__ tkeys, i64* __restrict__ tvals, u64* __restrict__ texm,

@@ -139,8 +139,15 @@ void tokenize_cache_spill(
     // balance at this grain).
     const char* v = getenv("MR_TOKENIZE_V5");
     const char* v4 = getenv("MR_TOKENIZE_V4");
-    if (!(v && v[0] == '1') && !(v4 && v4[0] == '1'))
-      hipLaunchKernelGGL(tokenize_v6_kernel,
+    if (!(v && v[0] == '1') && !(v4 && v4[0] == '1')) {
+      // MR_TOK_CACHE ∈ {512, 1024, 2048}: LDS-cache slots per block —
+      // occupancy (smaller cache -> more blocks/CU) vs spill volume
+      const char* cs = getenv("MR_TOK_CACHE");
+      int cache = cs ? atoi(cs) : 1024;
+      auto kfn = tokenize_v6_kernel<1024>;
+      if (cache == 512) kfn = tokenize_v6_kernel<512>;
+      else if (cache == 2048) kfn = tokenize_v6_kernel<2048>;
+      hipLaunchKernelGGL(kfn,
                          dim3(grid_for(n, TOK_BYTES)), dim3(kBlock), 0,
                          cur_stream(), text.data_ptr<u8>(), n, (u64)pos_base,
                          u64p(tkeys), tvals.data_ptr<i64>(),
@@ -149,7 +156,7 @@ void tokenize_cache_spill(
                          reinterpret_cast<unsigned long long*>(counter.data_ptr<i64>()),
                          spill_cap,
                          reinterpret_cast<unsigned long long*>(nwords.data_ptr<i64>()));
-    else if (!(v && v[0] == '1'))
+    } else if (!(v && v[0] == '1'))
       hipLaunchKernelGGL(tokenize_cache_spill_kernel,
                          dim3(grid_for(n, TOK_BYTES)), dim3(kBlock), 0,
                          cur_stream(), text.data_ptr<u8>(), n, (u64)pos_base,
