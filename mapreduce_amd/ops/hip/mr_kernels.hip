@@ -640,7 +640,7 @@ __global__ __launch_bounds__(256) void tokenize_v5_kernel(
 // extra barriers/LDS cost more than lane balance buys.
 // ---------------------------------------------------------------------------
 
-template <int TOK_CACHE>
+template <int CACHE_N>
 __global__ __launch_bounds__(256) void tokenize_v6_kernel(
     const u8* __restrict__ text, long n, u64 pos_base,
     u64* __restrict__ tkeys, i64* __restrict__ tvals, u64* __restrict__ texm,
@@ -648,10 +648,10 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
     unsigned long long* __restrict__ spill_counter, long spill_cap,
     unsigned long long* __restrict__ nwords) {
   __shared__ __align__(16) u8 tile[TOK_TILE + TOK_HALO];
-  __shared__ u64 ckeys[TOK_CACHE];
-  __shared__ u64 cpos[TOK_CACHE];
-  __shared__ u32 ccnt[TOK_CACHE];
-  for (int s = threadIdx.x; s < TOK_CACHE; s += blockDim.x) {
+  __shared__ u64 ckeys[CACHE_N];
+  __shared__ u64 cpos[CACHE_N];
+  __shared__ u32 ccnt[CACHE_N];
+  for (int s = threadIdx.x; s < CACHE_N; s += blockDim.x) {
     ckeys[s] = HT_EMPTY;
     ccnt[s] = 0;
   }
@@ -732,7 +732,7 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
       u64 k = remap_key(h);
       u64 p = ((pos_base + (u64)(base + my0 + s)) << 16) | (u64)len;
       // LDS cache insert (v4 path)
-      u32 slot = (u32)((k ^ (k >> 32)) & (TOK_CACHE - 1));
+      u32 slot = (u32)((k ^ (k >> 32)) & (CACHE_N - 1));
       bool done = false;
       for (int pr = 0; pr < TOK_PROBE; ++pr) {
         u64 cur2 = ckeys[slot];
@@ -752,7 +752,7 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
             break;
           }
         }
-        slot = (slot + 1) & (TOK_CACHE - 1);
+        slot = (slot + 1) & (CACHE_N - 1);
       }
       if (!done) {
         sh_[ns] = k;
@@ -770,7 +770,7 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
     }
   }
   __syncthreads();
-  for (int s = tid; s < TOK_CACHE; s += blockDim.x)
+  for (int s = tid; s < CACHE_N; s += blockDim.x)
     if (ckeys[s] != HT_EMPTY && ccnt[s])
       ht_add(ckeys[s], cpos[s], (i64)ccnt[s], tkeys, tvals, texm, cap_mask);
   unsigned long long ws = my_words;
