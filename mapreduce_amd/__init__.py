@@ -26,3 +26,32 @@ from .worker import Worker  # noqa: F401
 
 _NAME = "mapreduce_amd"
 _VERSION = "0.1.0"
+
+
+def utest() -> None:
+    """Package integrity self-check (init.lua:36-38 utest runner parity):
+    a quick in-process wordcount against the naive oracle."""
+    import collections
+
+    from .runner import run_local
+
+    counts = {}
+    data = {"1": "a b b c", "2": "b c c d d"}
+    fns = {
+        "taskfn": lambda emit: [emit(k, v) for k, v in data.items()],
+        "mapfn": lambda k, v, emit: [emit(w, 1) for w in v.split()],
+        "partitionfn": lambda k: len(k) % 3,
+        "reducefn": lambda k, vs, emit: emit(sum(vs)),
+        "finalfn": lambda pairs: counts.update(
+            {k: v[0] for k, v in pairs}) or True,
+        "associative_reducer": True,
+        "commutative_reducer": True,
+        "idempotent_reducer": True,
+    }
+    srv = run_local({"fns": {r: fns for r in (
+        "taskfn", "mapfn", "partitionfn", "reducefn", "finalfn")},
+        "verbose": False}, nworkers=2)
+    exp = collections.Counter(" ".join(data.values()).split())
+    assert counts == dict(exp), (counts, exp)
+    assert srv.finished
+    print("mapreduce_amd utest ok")

@@ -272,9 +272,11 @@ class Server:
                 self._log(f"iterative loop -> iteration {self.iteration}")
             else:
                 self.finished = True
-        self.task.set_task_status(TASK_STATUS.FINISHED)
-        self.task.drop_jobs()
         self.stats["total_time"] = gettime() - t_start
+        # persist the stats sub-document in the task singleton
+        # (server.lua:584-601 task:insert{stats})
+        self.task.set_task_status(TASK_STATUS.FINISHED, stats=self.stats)
+        self.task.drop_jobs()
         self.print_stats()
 
     def drop_all(self) -> None:

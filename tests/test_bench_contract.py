@@ -1,0 +1,63 @@
+"""bench.py contract tests: the driver runs
+`python -m torch.distributed.run --nnodes=1 --nproc-per-node N
+ --master-addr 127.0.0.1 bench.py --gpus N ...` — verify that exact
+invocation works (CPU device here; RCCL path shares all code but the
+backend)."""
+
+import json
+import os
+import socket
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+def run_bench(extra, env=None):
+    out = subprocess.run(
+        [sys.executable] + extra,
+        cwd=REPO, env=dict(os.environ, PYTHONPATH=REPO, **(env or {})),
+        capture_output=True, text=True, timeout=240)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    return json.loads(line)
+
+
+def test_bench_single_process_contract():
+    d = run_bench(["bench.py", "--steps", "2", "--warmup", "1",
+                   "--words", "20000", "--splits", "4", "--vocab", "500",
+                   "--device", "cpu"])
+    assert d["metric"] == "words/sec"
+    assert d["n_gpus"] == 1
+    assert d["steps"] == 2 and d["warmup"] == 1
+    assert d["higher_is_better"] is True
+    assert d["scaling"] == "weak"
+    assert d["value"] > 0 and d["ms_per_step"] > 0
+    assert d["vs_baseline"] == pytest.approx(
+        d["value"] / (49_158_635 / 49.23))
+    assert d["config"]["parallelism"] == "dp1"
+
+
+@pytest.mark.timeout(240)
+def test_bench_torchrun_ws2_contract():
+    port = free_port()
+    d = run_bench([
+        "-m", "torch.distributed.run", "--nnodes=1", "--nproc-per-node", "2",
+        "--master-addr", "127.0.0.1", "--master-port", str(port),
+        "bench.py", "--gpus", "2", "--steps", "2", "--warmup", "1",
+        "--words", "20000", "--splits", "4", "--vocab", "500",
+        "--device", "cpu"])
+    assert d["n_gpus"] == 2
+    assert d["config"]["parallelism"] == "dp2"
+    # whole-job aggregate: words per step = words x world
+    assert d["config"]["global_batch"] == 40000
