@@ -24,7 +24,7 @@ import pickle
 import torch
 
 _CFG = {"shards": 4, "iters": 3, "lr": 0.1, "bunch": 64, "seed": 0,
-        "hidden": 32, "cnn": None, "db": "mr"}
+        "hidden": 32, "cnn": None, "db": "mr", "device": "cpu"}
 STATE = {"model": None, "iteration": 0, "losses": [], "pt": None}
 
 
@@ -67,14 +67,14 @@ def _make_model():
     with torch.no_grad():
         for p in m.parameters():
             p.copy_(torch.randn(p.shape, generator=g) * 0.1)
-    return m
+    return m.to(_CFG["device"])
 
 
 def _shard_batch(shard: int, iteration: int):
     """Synthetic 8x8 'digits' batch, deterministic per (shard, iter)."""
     g = torch.Generator().manual_seed(1000 * iteration + shard)
-    x = torch.randn(_CFG["bunch"], 64, generator=g)
-    y = torch.randint(0, 10, (_CFG["bunch"],), generator=g)
+    x = torch.randn(_CFG["bunch"], 64, generator=g).to(_CFG["device"])
+    y = torch.randint(0, 10, (_CFG["bunch"],), generator=g).to(_CFG["device"])
     return x, y
 
 
@@ -87,7 +87,7 @@ def serialize_model(m) -> str:
 def deserialize_model(blob: str):
     m = _make_model()
     m.load_state_dict(torch.load(io.BytesIO(base64.b64decode(blob)),
-                                 weights_only=True))
+                                 weights_only=True, map_location="cpu"))
     return m
 
 
@@ -121,7 +121,9 @@ def mapfn(key, value, emit):
         x, y = _shard_batch(value["shard"], STATE["iteration"])
         loss = torch.nn.functional.cross_entropy(m(x), y)
         loss.backward()
-        grads = [(name, p.grad.detach().clone())
+        # emit on CPU: host-tier record streams pickle the values (the GPU
+        # gradient path proper is gpu/gradsum.py's bucketed allreduce)
+        grads = [(name, p.grad.detach().to("cpu"))
                  for name, p in m.named_parameters()]
     for name, g in grads:
         emit(name, g)
@@ -158,7 +160,7 @@ def finalfn(pairs):
     nshards = _CFG["shards"]
     with torch.no_grad():
         for name, p in m.named_parameters():
-            p -= _CFG["lr"] * grads[name] / nshards
+            p -= _CFG["lr"] * grads[name].to(p.device) / nshards
     STATE["losses"].append(float(loss[0] / loss[1]))
     STATE["iteration"] += 1
     pt = _pt()
