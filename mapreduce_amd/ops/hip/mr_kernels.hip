@@ -640,17 +640,21 @@ __global__ __launch_bounds__(256) void tokenize_v5_kernel(
 // extra barriers/LDS cost more than lane balance buys.
 // ---------------------------------------------------------------------------
 
-template <int CACHE_N>
+template <int CACHE_N, bool GPOS>
 __global__ __launch_bounds__(256) void tokenize_v6_kernel(
     const u8* __restrict__ text, long n, u64 pos_base,
     u64* __restrict__ tkeys, i64* __restrict__ tvals, u64* __restrict__ texm,
     u64 cap_mask, u64* __restrict__ out_hash, u64* __restrict__ out_pos,
     unsigned long long* __restrict__ spill_counter, long spill_cap,
-    unsigned long long* __restrict__ nwords) {
+    unsigned long long* __restrict__ nwords, u64* __restrict__ cpos_g) {
   __shared__ __align__(16) u8 tile[TOK_TILE + TOK_HALO];
   __shared__ u64 ckeys[CACHE_N];
-  __shared__ u64 cpos[CACHE_N];
+  // exemplar positions: LDS normally; with GPOS a global side-buffer
+  // (written once per distinct word per block — off the critical path)
+  // frees 16 KB LDS for occupancy
+  __shared__ u64 cpos_l[GPOS ? 1 : CACHE_N];
   __shared__ u32 ccnt[CACHE_N];
+  u64* cpos = GPOS ? &cpos_g[(u64)blockIdx.x * CACHE_N] : cpos_l;
   for (int s = threadIdx.x; s < CACHE_N; s += blockDim.x) {
     ckeys[s] = HT_EMPTY;
     ccnt[s] = 0;

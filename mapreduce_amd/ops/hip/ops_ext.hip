@@ -145,18 +145,34 @@ void tokenize_cache_spill(
       int cache = cs ? atoi(cs) : 2048;  // sweep: 512=5.52, 1024=5.26,
                                          // 2048=4.94 ms/step — fewer
                                          // spills beat occupancy here
-      auto kfn = tokenize_v6_kernel<2048>;
-      if (cache == 512) kfn = tokenize_v6_kernel<512>;
-      else if (cache == 1024) kfn = tokenize_v6_kernel<1024>;
+      const char* gp = getenv("MR_TOK_GPOS");
+      bool gpos = gp && gp[0] == '1';
+      auto kfn = tokenize_v6_kernel<2048, false>;
+      if (gpos) kfn = tokenize_v6_kernel<2048, true>;
+      else if (cache == 512) kfn = tokenize_v6_kernel<512, false>;
+      else if (cache == 1024) kfn = tokenize_v6_kernel<1024, false>;
+      long blocks = grid_for(n, TOK_BYTES);
+      static torch::Tensor cpos_g;  // persistent side-buffer (GPOS only)
+      u64* cpg = nullptr;
+      if (gpos) {
+        long need = blocks * 2048;
+        if (!cpos_g.defined() || cpos_g.numel() < need ||
+            cpos_g.device() != text.device())
+          cpos_g = torch::empty({need},
+                                torch::TensorOptions().device(text.device())
+                                    .dtype(torch::kInt64));
+        cpg = u64p(cpos_g);
+      }
       hipLaunchKernelGGL(kfn,
-                         dim3(grid_for(n, TOK_BYTES)), dim3(kBlock), 0,
+                         dim3(blocks), dim3(kBlock), 0,
                          cur_stream(), text.data_ptr<u8>(), n, (u64)pos_base,
                          u64p(tkeys), tvals.data_ptr<i64>(),
                          texm.numel() ? u64p(texm) : nullptr, (u64)(cap - 1),
                          u64p(out_hash), u64p(out_pos),
                          reinterpret_cast<unsigned long long*>(counter.data_ptr<i64>()),
                          spill_cap,
-                         reinterpret_cast<unsigned long long*>(nwords.data_ptr<i64>()));
+                         reinterpret_cast<unsigned long long*>(nwords.data_ptr<i64>()),
+                         cpg);
     } else if (!(v && v[0] == '1'))
       hipLaunchKernelGGL(tokenize_cache_spill_kernel,
                          dim3(grid_for(n, TOK_BYTES)), dim3(kBlock), 0,
