@@ -62,8 +62,19 @@ def exchange(data: torch.Tensor, send_counts: List[int],
     rank, world = world_info(group)
     if world == 1:
         return data.clone()
-    out = torch.empty(int(sum(recv_counts)), dtype=data.dtype,
-                      device=data.device)
+    need = int(sum(recv_counts))
+    if data.is_cuda:
+        # shuffle-skew memory guard (SURVEY.md §7 hard parts): a hot
+        # partition must not silently OOM the rank — fail with an
+        # actionable message instead (re-partition finer / spill to host)
+        free, _ = torch.cuda.mem_get_info(data.device)
+        if need * data.element_size() > free * 0.9:
+            raise RuntimeError(
+                f"rank {rank}: all-to-all receive of "
+                f"{need * data.element_size() / 1e9:.1f} GB exceeds free "
+                f"HBM ({free / 1e9:.1f} GB) — partition skew; increase the "
+                "partition count or enable host spill")
+    out = torch.empty(need, dtype=data.dtype, device=data.device)
     try:
         dist.all_to_all_single(out, data.contiguous(),
                                output_split_sizes=recv_counts,
