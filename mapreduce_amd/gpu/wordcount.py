@@ -48,8 +48,15 @@ class WordCountResult:
         lens, blob = ops.extract_words(self.blob_src, self.pos)
         return (self.keys.cpu(), self.counts.cpu(), lens.cpu(), blob.cpu())
 
-    def to_host(self) -> List[Tuple[bytes, int]]:
-        """Materialize (word, count) pairs, sorted by key hash (C8)."""
+    def to_host(self, order: str = "hash") -> List[Tuple[bytes, int]]:
+        """Materialize (word, count) pairs (C8).
+
+        order="hash" (default): the engine's native u64-hash order —
+        deterministic and grouped, the order the device arrays are in.
+        order="lex": lexicographic by word bytes, matching the reference's
+        sorted-result guarantee (job.lua:194, server.lua:360-385) — a
+        host-side sort of the (small) unique set at the finalfn boundary,
+        where the reference also pays its string costs."""
         lens, blob = ops.extract_words(self.blob_src, self.pos)
         raw = bytes(blob.cpu().numpy().tobytes())
         counts = self.counts.cpu().tolist()
@@ -58,6 +65,8 @@ class WordCountResult:
         for L, c in zip(lens.cpu().tolist(), counts):
             out.append((raw[off:off + L], c))
             off += L
+        if order == "lex":
+            out.sort(key=lambda kv: kv[0])
         return out
 
 

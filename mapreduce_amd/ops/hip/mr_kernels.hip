@@ -714,24 +714,41 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
       u32 len;
       if (t) {
         len = (u32)(__ffs(t) - 1);  // bit s is clear, so len >= 1
-        // hash bytes s..s+len-1 from q0..q3 (s+len <= 32 by construction):
-        // branchless 4-way register select per byte — no LDS re-reads
-        for (u32 b2 = 0; b2 < len; ++b2) {
-          int j = s + (int)b2;
-          u64 lo = (j & 16) ? q2 : q0;
-          u64 hi = (j & 16) ? q3 : q1;
-          u64 w = (j & 8) ? hi : lo;
-          h = (h ^ ((w >> (8 * (j & 7))) & 0xFF)) * FNV64_PRIME;
+        // chunked hash from registers: bytes s..s+len-1 of q0..q3 as
+        // 8-byte windows (branchless selects, zero LDS re-reads; the
+        // last window masks to the remaining bytes)
+        for (u32 c = 0; c < len; c += 8) {
+          int j0 = s + (int)c;
+          int qi = j0 >> 3;  // 0..4 (window may straddle past byte 31)
+          u64 lo = (qi & 2) ? ((qi & 1) ? q3 : q2) : ((qi & 1) ? q1 : q0);
+          u64 hi = (qi >= 3) ? 0 : ((qi & 2) ? q3 : ((qi & 1) ? q2 : q1));
+          int sh2 = 8 * (j0 & 7);
+          u64 w = sh2 ? ((lo >> sh2) | (hi << (64 - sh2))) : lo;
+          u32 rem = len - c;
+          if (rem < 8) w &= (((u64)1 << (8 * rem)) - 1);
+          h = whash_chunk(h, w);
         }
+        h = whash_fin(h, len);
       } else {
-        // word longer than the 32-byte window: scan from global
+        // word longer than the 32-byte window: scan from global,
+        // accumulating the same 8-byte chunks
         long g = base + my0 + s;
+        u64 chunk = 0;
+        int cb = 0;
+        long wlen = 0;
         while (g < n && !is_ws(text[g])) {
-          h = (h ^ text[g]) * FNV64_PRIME;
+          chunk |= (u64)text[g] << (8 * cb);
+          if (++cb == 8) {
+            h = whash_chunk(h, chunk);
+            chunk = 0;
+            cb = 0;
+          }
           ++g;
+          ++wlen;
         }
-        len = (u32)((g - (base + my0 + s)) > 0xFFFF
-                        ? 0xFFFF : g - (base + my0 + s));
+        if (cb) h = whash_chunk(h, chunk);
+        h = whash_fin(h, (u64)wlen);
+        len = (u32)(wlen > 0xFFFF ? 0xFFFF : wlen);
       }
       u64 k = remap_key(h);
       u64 p = ((pos_base + (u64)(base + my0 + s)) << 16) | (u64)len;

@@ -68,6 +68,21 @@ def fnv1a64(data: Any) -> int:
     return h
 
 
+def wordhash64(data: Any) -> int:
+    """Chunked 64-bit word hash: one xor-multiply per 8 little-endian
+    bytes (zero-padded) + a length fold.  The GPU engine's dictionary hash
+    (tokenize_v6): 6x fewer dependent multiplies than byte-serial FNV-1a
+    and provably collision-free for distinct words of <= 8 bytes (the
+    multiply by an odd prime is bijective).  Must match whash_* in
+    ops/hip/common.h."""
+    b = _key_bytes(data)
+    h = FNV64_OFFSET
+    for i in range(0, len(b), 8):
+        chunk = int.from_bytes(b[i:i + 8], "little")
+        h = ((h ^ chunk) * FNV64_PRIME) & _MASK64
+    return ((h ^ len(b)) * FNV64_PRIME) & _MASK64
+
+
 def jenkins_oaat(data: Any) -> int:
     """Jenkins one-at-a-time 32-bit hash (tuple.lua:121-140)."""
     h = 0

@@ -227,6 +227,14 @@ def test_streaming_wordcount_mode_vs_counter(dev):
     got = dict(res.to_host())
     exp = collections.Counter(vocab[i] for i in widx.tolist())
     assert got == dict(exp)
+    # hash parity: the engine's dictionary hash must match
+    # utils.tuple.wordhash64 bit-for-bit (partitioning consistency)
+    from mapreduce_amd.utils.tuple import wordhash64
+    got_keys = set(u64view(res.keys.cpu()).tolist())
+    assert got_keys == {wordhash64(w) for w in exp}
+    # lexicographic materialization option
+    lex = res.to_host(order="lex")
+    assert [w for w, _ in lex] == sorted(got)
     # A/B: fused mode must produce identical counts
     job2 = WordCountJob(dev, vocab_estimate=6000, mode="fused")
     res2 = job2.run(text)
