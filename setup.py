@@ -7,8 +7,16 @@ to GPU boxes (no JIT cache involved).
 """
 
 import os
+import shutil
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+# The ninja rule for hipcc emits no depfile, so edits to #include'd .hip
+# kernels do NOT trigger a recompile (a stale .so once shipped an old
+# kernel).  The extension is one TU (~90 s): always build fresh.
+shutil.rmtree(os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                           "build", "temp.linux-x86_64-3.10"),
+              ignore_errors=True)
 
 from setuptools import setup
 from torch.utils.cpp_extension import BuildExtension, CUDAExtension
