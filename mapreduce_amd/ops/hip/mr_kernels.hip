@@ -705,12 +705,7 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
     const u64 q3 = (u64)vb.z | ((u64)vb.w << 32);
     u64 sh_[8];
     u64 sp_[8];
-    u64 wk_[8];
-    u64 wp_[8];
-    u64 wcur_[8];
-    u32 wslot_[8];
     int ns = 0;
-    int nw2 = 0;
     while (sm) {
       int s = __ffs(sm) - 1;
       sm &= sm - 1;
@@ -757,25 +752,11 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
       }
       u64 k = remap_key(h);
       u64 p = ((pos_base + (u64)(base + my0 + s)) << 16) | (u64)len;
-      // collect the word and ISSUE its first probe load now — the
-      // ckeys read of word i overlaps the hash compute of word i+1
-      // (the wait lands at first use in the insert loop below), cutting
-      // the serial LDS-latency chain per word
-      u32 slot0 = (u32)((k ^ (k >> 32)) & (CACHE_N - 1));
-      wk_[nw2] = k;
-      wp_[nw2] = p;
-      wslot_[nw2] = slot0;
-      wcur_[nw2] = ckeys[slot0];
-      ++nw2;
-    }
-    // ---- insert pass (first probes pre-resolved)
-    for (int i = 0; i < nw2; ++i) {
-      u64 k = wk_[i];
-      u64 p = wp_[i];
-      u32 slot = wslot_[i];
-      u64 cur2 = wcur_[i];
+      // LDS cache insert (v4 path)
+      u32 slot = (u32)((k ^ (k >> 32)) & (CACHE_N - 1));
       bool done = false;
       for (int pr = 0; pr < TOK_PROBE; ++pr) {
+        u64 cur2 = ckeys[slot];
         if (cur2 == k) {
           atomicAdd(&ccnt[slot], 1u);
           done = true;
@@ -791,12 +772,8 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
             done = true;
             break;
           }
-          // CAS lost to a different key: fall through and re-probe
-          cur2 = ckeys[slot];
-          continue;
         }
         slot = (slot + 1) & (CACHE_N - 1);
-        cur2 = ckeys[slot];
       }
       if (!done) {
         sh_[ns] = k;
