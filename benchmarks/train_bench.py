@@ -1,0 +1,68 @@
+"""Gradient-training MapReduce benchmark (BASELINE.json config 5: the
+cnn.lua MNIST-digits iterative training analogue).
+
+Measures full MapReduce iterations/second of examples.train_digits through
+the real scheduler (threaded workers, host-tier reduce; model compute on
+GPU when available).
+
+    python benchmarks/train_bench.py --iters 10 --shards 8 --workers 4
+"""
+
+from __future__ import annotations
+
+import argparse
+import importlib
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--iters", type=int, default=10)
+    p.add_argument("--shards", type=int, default=8)
+    p.add_argument("--workers", type=int, default=4)
+    p.add_argument("--bunch", type=int, default=256)
+    p.add_argument("--hidden", type=int, default=128)
+    args = p.parse_args()
+
+    import mapreduce_amd.examples.train_digits as td
+    from mapreduce_amd import job as jobmod, run_local
+    importlib.reload(td)
+    jobmod._module_cache.clear()
+    jobmod._inited.clear()
+
+    device = "cuda:0" if torch.cuda.is_available() else "cpu"
+    fns = {r: td for r in ("taskfn", "mapfn", "partitionfn", "reducefn",
+                           "combinerfn", "finalfn")}
+    t0 = time.perf_counter()
+    srv = run_local({"fns": fns, "verbose": False,
+                     "init_args": {"shards": args.shards,
+                                   "iters": args.iters, "lr": 0.05,
+                                   "bunch": args.bunch,
+                                   "hidden": args.hidden,
+                                   "device": device}},
+                    nworkers=args.workers)
+    el = time.perf_counter() - t0
+    assert srv.finished and len(td.STATE["losses"]) == args.iters
+    print(json.dumps({
+        "metric": "training iterations/sec (full MapReduce loop)",
+        "value": args.iters / el,
+        "unit": "iters/s",
+        "iters": args.iters,
+        "shards": args.shards,
+        "workers": args.workers,
+        "device": device,
+        "losses": td.STATE["losses"],
+        "higher_is_better": True,
+        "data": "synthetic digits batches, random-init MLP",
+    }), flush=True)
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -127,7 +127,7 @@ def mapfn(key, value, emit):
                  for name, p in m.named_parameters()]
     for name, g in grads:
         emit(name, g)
-    emit("__loss__", torch.tensor([float(loss), 1.0]))
+    emit("__loss__", torch.tensor([float(loss.detach()), 1.0]))
 
 
 def partitionfn(key):

@@ -640,14 +640,14 @@ __global__ __launch_bounds__(256) void tokenize_v5_kernel(
 // extra barriers/LDS cost more than lane balance buys.
 // ---------------------------------------------------------------------------
 
-template <int CACHE_N, bool GPOS>
+template <int CACHE_N, bool GPOS, int TILE_N>
 __global__ __launch_bounds__(256) void tokenize_v6_kernel(
     const u8* __restrict__ text, long n, u64 pos_base,
     u64* __restrict__ tkeys, i64* __restrict__ tvals, u64* __restrict__ texm,
     u64 cap_mask, u64* __restrict__ out_hash, u64* __restrict__ out_pos,
     unsigned long long* __restrict__ spill_counter, long spill_cap,
     unsigned long long* __restrict__ nwords, u64* __restrict__ cpos_g) {
-  __shared__ __align__(16) u8 tile[TOK_TILE + TOK_HALO];
+  __shared__ __align__(16) u8 tile[TILE_N + TOK_HALO];
   __shared__ u64 ckeys[CACHE_N];
   // exemplar positions: LDS normally; with GPOS a global side-buffer
   // (written once per distinct word per block — off the critical path)
@@ -661,12 +661,12 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
   }
   const int tid = threadIdx.x;
   unsigned long long my_words = 0;
-  long tile0 = (long)blockIdx.x * TOK_TILE;
-  long tstride = (long)gridDim.x * TOK_TILE;
+  long tile0 = (long)blockIdx.x * TILE_N;
+  long tstride = (long)gridDim.x * TILE_N;
   for (long base = tile0; base < n; base += tstride) {
     __syncthreads();
     long avail = n - base;
-    long want = avail < TOK_TILE + TOK_HALO ? avail : TOK_TILE + TOK_HALO;
+    long want = avail < TILE_N + TOK_HALO ? avail : TILE_N + TOK_HALO;
     for (int o = tid * 16; o < want; o += blockDim.x * 16) {
       if (o + 16 <= want && (((uintptr_t)&text[base + o]) & 15) == 0) {
         *(uint4*)&tile[o] = *(const uint4*)&text[base + o];
@@ -676,8 +676,9 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
       }
     }
     __syncthreads();
-    long my0 = (long)tid * TOK_BYTES;
-    if (my0 >= avail) continue;
+    for (int wnd = 0; wnd < TILE_N / 4096; ++wnd) {
+    long my0 = ((long)tid + (long)wnd * 256) * TOK_BYTES;
+    if (my0 >= avail) break;
     // 32 bytes in registers: my window + 16B lookahead (halo-staged)
     const uint4 va = *(const uint4*)&tile[my0];
     const uint4 vb = *(const uint4*)&tile[my0 + 16];
@@ -789,6 +790,7 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
           out_pos[o + w] = sp_[w];
         }
     }
+    }  // wnd
   }
   __syncthreads();
   for (int s = tid; s < CACHE_N; s += blockDim.x)

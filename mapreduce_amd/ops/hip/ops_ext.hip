@@ -147,11 +147,14 @@ void tokenize_cache_spill(
                                          // spills beat occupancy here
       const char* gp = getenv("MR_TOK_GPOS");
       bool gpos = !(gp && gp[0] == '0');  // default ON: 4.90 vs 5.05 ms
-      auto kfn = tokenize_v6_kernel<2048, false>;
-      if (gpos) kfn = tokenize_v6_kernel<2048, true>;
-      else if (cache == 512) kfn = tokenize_v6_kernel<512, false>;
-      else if (cache == 1024) kfn = tokenize_v6_kernel<1024, false>;
-      long blocks = grid_for(n, TOK_BYTES);
+      const char* ts = getenv("MR_TOK_TILE");
+      int tsz = ts ? atoi(ts) : 4096;
+      auto kfn = tokenize_v6_kernel<2048, false, 4096>;
+      if (gpos && tsz == 8192) kfn = tokenize_v6_kernel<2048, true, 8192>;
+      else if (gpos) kfn = tokenize_v6_kernel<2048, true, 4096>;
+      else if (cache == 512) kfn = tokenize_v6_kernel<512, false, 4096>;
+      else if (cache == 1024) kfn = tokenize_v6_kernel<1024, false, 4096>;
+      long blocks = grid_for(n, TOK_BYTES * (tsz / 4096));
       static torch::Tensor cpos_g;  // persistent side-buffer (GPOS only)
       u64* cpg = nullptr;
       if (gpos) {
