@@ -129,6 +129,24 @@ class GpuClusterRunner:
                                     real_time=gettime()
                                     - claimed["started_time"]))
 
+    def _map_phase_with_retry(self, text, splits, attempts: int = 2):
+        """GPU map jobs append into shared device state (table + spill
+        stream), so a mid-phase failure cannot be retried at job
+        granularity without double-counting — retry is PHASE-scoped: reset
+        the device state and re-run every local job (all jobs are
+        rank-affine and deterministic, so the replay is exact).  This is
+        the HBM analogue of the reference's idempotent
+        remove_file-before-write republish (job.lua:219)."""
+        for attempt in range(attempts):
+            try:
+                self.job.begin_map(text)
+                self._run_map_jobs(splits)
+                return
+            except Exception:
+                if attempt + 1 >= attempts:
+                    raise
+                self._insert_map_jobs(splits)  # re-arm job docs
+
     # --------------------------------------------------------------- run
     def run(self, text: torch.Tensor, splits: List[Tuple[int, int]]):
         """One MapReduce job under control-plane tracking.  Returns the
@@ -138,11 +156,10 @@ class GpuClusterRunner:
                 "fns": {"engine": type(self.job).__name__},
                 "storage": "hbm", "result_ns": "result",
             }, 1)
-        self.job.begin_map(text)
         self._insert_map_jobs(splits)
         if self.rank == 0:
             self.task.set_task_status(TASK_STATUS.MAP)
-        self._run_map_jobs(splits)
+        self._map_phase_with_retry(text, splits)
         # local jobs all WRITTEN; the barrier is the cross-rank "all maps
         # done" agreement (C4 as a collective instead of a DB poll)
         dx.barrier(self.group)

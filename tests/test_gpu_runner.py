@@ -32,6 +32,30 @@ def test_runner_single_rank(claim_mode):
         assert st["status"] == STATUS.WRITTEN
 
 
+def test_runner_phase_retry_after_transient_fault():
+    """A transient map-split failure triggers a PHASE-scoped retry (device
+    state reset + exact replay) and still yields correct counts."""
+    c = make_corpus("cpu", nwords=4000, nsplits=4, vocab_size=200, seed=9)
+
+    class FlakyJob(WordCountJob):
+        def __init__(self, *a, **k):
+            super().__init__(*a, **k)
+            self.calls = 0
+
+        def map_split(self, s, e):
+            self.calls += 1
+            if self.calls == 2:
+                raise RuntimeError("injected transient device fault")
+            super().map_split(s, e)
+
+    job = FlakyJob("cpu", vocab_estimate=400)
+    runner = GpuClusterRunner(job, claim_mode="dynamic")
+    res = runner.run(c.text, c.splits())
+    assert res.nwords == 4000
+    assert dict(res.to_host()) == oracle(bytes(c.text.numpy().tobytes()))
+    assert job.calls >= 5  # first attempt partial + full replay
+
+
 def _worker(rank, world, port, claim_mode):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
