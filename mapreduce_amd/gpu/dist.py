@@ -45,14 +45,19 @@ def world_info(group=None) -> Tuple[int, int]:
 
 
 def exchange_counts(send_counts: torch.Tensor, group=None) -> torch.Tensor:
-    """recv_counts[i] = send_counts of rank i toward me (C5 size exchange).
-    send_counts: i64[world] on the comm device."""
+    """Size exchange for the all-to-all (C5): send_counts is i64[k*world]
+    (k independent segments packed back-to-back; callers pack several
+    count arrays into ONE collective).  Returns recv of the same shape
+    with recv[s*world + i] = rank i's send_counts[s*world + rank]."""
     rank, world = world_info(group)
     if world == 1:
         return send_counts.clone()
+    assert send_counts.numel() % world == 0
+    segs = send_counts.numel() // world
     mat = [torch.zeros_like(send_counts) for _ in range(world)]
     dist.all_gather(mat, send_counts.contiguous(), group=group)
-    return torch.stack([mat[i][rank] for i in range(world)])
+    stacked = torch.stack(mat).view(world, segs, world)
+    return stacked[:, :, rank].transpose(0, 1).reshape(-1)
 
 
 def exchange(data: torch.Tensor, send_counts: List[int],

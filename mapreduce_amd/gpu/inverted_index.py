@@ -133,10 +133,13 @@ class InvertedIndexJob:
                 blob_counts_d = torch.cat([cum[:1], cum[1:] - cum[:-1]])
             else:
                 blob_counts_d = torch.zeros_like(counts_d)
-            recv_counts = dx.exchange_counts(counts_d, self.group)
-            recv_blob = dx.exchange_counts(blob_counts_d, self.group)
-            sc, rc = counts_d.cpu().tolist(), recv_counts.cpu().tolist()
-            sb, rb = blob_counts_d.cpu().tolist(), recv_blob.cpu().tolist()
+            packed = torch.cat([counts_d, blob_counts_d])
+            recv_packed = dx.exchange_counts(packed, self.group)
+            host = torch.stack([packed, recv_packed]).cpu()
+            sc = host[0, :self.world].tolist()
+            sb = host[0, self.world:].tolist()
+            rc = host[1, :self.world].tolist()
+            rb = host[1, self.world:].tolist()
             rh = dx.exchange(uh, sc, rc, self.group)
             rd = dx.exchange(ud, sc, rc, self.group)
             rtf = dx.exchange(tf, sc, rc, self.group)

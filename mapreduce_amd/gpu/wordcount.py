@@ -198,12 +198,16 @@ class WordCountJob:
                 blob_counts_d = torch.cat([cum[:1], cum[1:] - cum[:-1]])
             else:
                 blob_counts_d = torch.zeros_like(counts_d)
-            recv_counts_d = dx.exchange_counts(counts_d, self.group)
-            recv_blob_d = dx.exchange_counts(blob_counts_d, self.group)
-            send_c = counts_d.cpu().tolist()
-            recv_c = recv_counts_d.cpu().tolist()
-            send_b = blob_counts_d.cpu().tolist()
-            recv_b = recv_blob_d.cpu().tolist()
+            # ONE packed size exchange + ONE host sync for both arrays
+            # (xGMI collectives and D2H syncs are per-call latency-bound;
+            # fewer+larger wins — guide)
+            packed = torch.cat([counts_d, blob_counts_d])
+            recv_packed = dx.exchange_counts(packed, self.group)
+            host = torch.stack([packed, recv_packed]).cpu()
+            send_c = host[0, :self.world].tolist()
+            send_b = host[0, self.world:].tolist()
+            recv_c = host[1, :self.world].tolist()
+            recv_b = host[1, self.world:].tolist()
             rk = dx.exchange(sk, send_c, recv_c, self.group)
             rv = dx.exchange(sv, send_c, recv_c, self.group)
             rlens = dx.exchange(lens, send_c, recv_c, self.group)
