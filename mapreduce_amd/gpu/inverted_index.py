@@ -140,10 +140,14 @@ class InvertedIndexJob:
             sb = host[0, self.world:].tolist()
             rc = host[1, :self.world].tolist()
             rb = host[1, self.world:].tolist()
-            rh = dx.exchange(uh, sc, rc, self.group)
-            rd = dx.exchange(ud, sc, rc, self.group)
-            rtf = dx.exchange(tf, sc, rc, self.group)
-            rlens = dx.exchange(lens, sc, rc, self.group)
+            quad = torch.stack([uh, ud, tf, lens], dim=1).reshape(-1)
+            rquad = dx.exchange(quad, [4 * c for c in sc],
+                                [4 * c for c in rc], self.group)
+            rquad = rquad.view(-1, 4)
+            rh = rquad[:, 0].contiguous()
+            rd = rquad[:, 1].contiguous()
+            rtf = rquad[:, 2].contiguous()
+            rlens = rquad[:, 3].contiguous()
             rblob = dx.exchange(blob, sb, rb, self.group)
             roff = torch.cumsum(rlens, 0) - rlens
             rp = (roff << 16) | rlens

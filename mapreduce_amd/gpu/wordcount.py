@@ -208,9 +208,16 @@ class WordCountJob:
             send_b = host[0, self.world:].tolist()
             recv_c = host[1, :self.world].tolist()
             recv_b = host[1, self.world:].tolist()
-            rk = dx.exchange(sk, send_c, recv_c, self.group)
-            rv = dx.exchange(sv, send_c, recv_c, self.group)
-            rlens = dx.exchange(lens, send_c, recv_c, self.group)
+            # ONE i64 all-to-all for (key, count, len): rows interleave the
+            # three arrays, so each partition's segment stays contiguous
+            # and split sizes just triple
+            tri = torch.stack([sk, sv, lens], dim=1).reshape(-1)
+            rtri = dx.exchange(tri, [3 * c for c in send_c],
+                               [3 * c for c in recv_c], self.group)
+            rtri = rtri.view(-1, 3)
+            rk = rtri[:, 0].contiguous()
+            rv = rtri[:, 1].contiguous()
+            rlens = rtri[:, 2].contiguous()
             rblob = dx.exchange(blob, send_b, recv_b, self.group)
             # rebuild packed positions into the received blob
             roff = torch.cumsum(rlens, 0) - rlens
