@@ -79,7 +79,7 @@ class WordCountJob:
         as the low-memory path)."""
 
     def __init__(self, device, vocab_estimate: int = 1 << 18, group=None,
-                 mode: str = "auto"):
+                 mode: str = "auto", timing: bool = False):
         self.device = torch.device(device)
         self.group = group
         self.rank, self.world = dx.world_info(group)
@@ -89,6 +89,27 @@ class WordCountJob:
         self.mode = mode
         self.table = ops.make_table(vocab_estimate, self.device)
         self._nwords = torch.zeros(1, dtype=torch.int64, device=self.device)
+        # per-phase tracing (the job-document timestamps of job.lua:117-152
+        # in HIP-event form; stats format parity with server.lua:557-602)
+        self.timing = timing and self.device.type == "cuda"
+        self.last_phase_ms: dict = {}
+        self._events: list = []
+
+    def _mark(self, name: str) -> None:
+        if self.timing:
+            ev = torch.cuda.Event(enable_timing=True)
+            ev.record()
+            self._events.append((name, ev))
+
+    def _collect_timing(self) -> None:
+        if not self.timing or len(self._events) < 2:
+            return
+        self._events[-1][1].synchronize()
+        out = {}
+        for (n0, e0), (n1, e1) in zip(self._events, self._events[1:]):
+            out[n1] = e0.elapsed_time(e1)
+        self.last_phase_ms = out
+        self._events = []
 
     def reset(self, vocab_estimate: int = 0):
         self.table = ops.make_table(vocab_estimate or self.vocab_estimate,
@@ -166,13 +187,19 @@ class WordCountJob:
         # Split boundaries are whitespace-aligned, so tokenization over the
         # coalesced range is byte-identical to per-split runs.
         self.begin_map(text)
+        self._mark("start")
         if self._coalesced(text, splits):
             self.map_split(splits[0][0], splits[-1][1])
         else:
             for (s, e) in splits:
                 self.map_split(s, e)
+        self._mark("map_tokenize")
         nwords = self.finish_map()
-        return self.shuffle_reduce(nwords)
+        self._mark("map_combine")
+        res = self.shuffle_reduce(nwords)
+        self._mark("shuffle_reduce")
+        self._collect_timing()
+        return res
 
     def shuffle_reduce(self, nwords: int) -> WordCountResult:
         """Phase 2: extract + sort uniques, all-to-all exchange, segmented

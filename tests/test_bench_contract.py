@@ -48,6 +48,20 @@ def test_bench_single_process_contract():
     assert d["config"]["parallelism"] == "dp1"
 
 
+@pytest.mark.timeout(300)
+def test_bench_torchrun_ws4_contract():
+    """4-rank canary for the driver's N=4/8 scaling runs."""
+    port = free_port()
+    d = run_bench([
+        "-m", "torch.distributed.run", "--nnodes=1", "--nproc-per-node", "4",
+        "--master-addr", "127.0.0.1", "--master-port", str(port),
+        "bench.py", "--gpus", "4", "--steps", "2", "--warmup", "1",
+        "--words", "12000", "--splits", "3", "--vocab", "400",
+        "--device", "cpu"])
+    assert d["n_gpus"] == 4
+    assert d["config"]["global_batch"] == 48000
+
+
 @pytest.mark.timeout(240)
 def test_bench_torchrun_ws2_contract():
     port = free_port()
