@@ -241,6 +241,35 @@ def test_streaming_wordcount_mode_vs_counter(dev):
     assert dict(res2.to_host()) == got
 
 
+def test_streamed_file_wordcount_gpu(dev, tmp_path):
+    """K8 path on hardware: pinned staging + side-stream H2D chunks feed
+    the engine; counts equal the file oracle; phase tracing populates."""
+    from mapreduce_amd.gpu.input import StreamLoader, load_corpus
+    from mapreduce_amd.gpu.wordcount import WordCountJob
+    rng = np.random.default_rng(41)
+    p = tmp_path / "in.txt"
+    words = [f"s{int(x)}" for x in rng.integers(0, 500, size=200_000)]
+    p.write_text(" ".join(words))
+    exp = collections.Counter(open(p, "rb").read().split())
+
+    got = collections.Counter()
+    job = WordCountJob(dev, vocab_estimate=1000, timing=True)
+    total = 0
+    for chunk, base in StreamLoader(str(p), dev, chunk_bytes=256 << 10):
+        res = job.run(chunk)
+        total += res.nwords
+        for w, n in res.to_host():
+            got[w] += n
+    assert total == len(words)
+    assert got == exp
+    assert job.last_phase_ms and "map_tokenize" in job.last_phase_ms
+
+    # one-shot load_corpus path
+    c = load_corpus([str(p)], dev)
+    res = job.run(c.text, c.splits())
+    assert dict(res.to_host()) == dict(exp)
+
+
 def test_gpu_wordcount_pipeline_vs_counter(dev):
     """Fused single-GPU wordcount: tokenize -> hash combine -> sort uniques
     -> counts, vs collections.Counter (the naive oracle)."""
