@@ -640,7 +640,8 @@ __global__ __launch_bounds__(256) void tokenize_v5_kernel(
 // extra barriers/LDS cost more than lane balance buys.
 // ---------------------------------------------------------------------------
 
-template <int CACHE_N, bool GPOS, int TILE_N>
+// MODE (ablation, §5.4 rule 8): 0=full, 1=stage+classify, 2=+hash, 3=+cache (no spill)
+template <int CACHE_N, bool GPOS, int TILE_N, int MODE = 0>
 __global__ __launch_bounds__(256) void tokenize_v6_kernel(
     const u8* __restrict__ text, long n, u64 pos_base,
     u64* __restrict__ tkeys, i64* __restrict__ tvals, u64* __restrict__ texm,
@@ -700,6 +701,7 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
     long limw = avail - my0;  // starts must be real data bytes
     if (limw < 16) sm &= (limw <= 0) ? 0u : ((1u << limw) - 1);
     my_words += __popc(sm);
+    if (MODE == 1) { my_words += m32; continue; }
     const u64 q0 = (u64)va.x | ((u64)va.y << 32);
     const u64 q1 = (u64)va.z | ((u64)va.w << 32);
     const u64 q2 = (u64)vb.x | ((u64)vb.y << 32);
@@ -751,6 +753,7 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
         h = whash_fin(h, (u64)wlen);
         len = (u32)(wlen > 0xFFFF ? 0xFFFF : wlen);
       }
+      if (MODE == 2) { my_words += h; continue; }
       u64 k = remap_key(h);
       u64 p = ((pos_base + (u64)(base + my0 + s)) << 16) | (u64)len;
       // LDS cache insert (v4 path)
@@ -782,6 +785,7 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
         ++ns;
       }
     }
+    if (MODE == 3) { my_words += ns; ns = 0; }
     if (ns) {
       unsigned long long o = atomicAdd(spill_counter, (unsigned long long)ns);
       for (int w = 0; w < ns; ++w)

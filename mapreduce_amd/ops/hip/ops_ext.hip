@@ -199,6 +199,54 @@ void tokenize_cache_spill(
   }
 }
 
+// v6-structure ablation: mode 1=stage+classify, 2=+hash, 3=+cache, 0=full
+double tok6_ablate(torch::Tensor text, long mode, long iters) {
+  long n = text.numel();
+  auto opts = torch::TensorOptions().device(text.device()).dtype(torch::kInt64);
+  long tcap = 1 << 19;
+  auto tkeys = torch::full({tcap}, -1, opts);
+  auto tvals = torch::zeros({tcap}, opts);
+  auto texm = torch::zeros({tcap}, opts);
+  long cap = n / 2 + 16;
+  auto oh = torch::empty({cap}, opts);
+  auto op = torch::empty({cap}, opts);
+  auto ctr = torch::zeros({1}, opts);
+  auto nw = torch::zeros({1}, opts);
+  long blocks = grid_for(n, TOK_BYTES);
+  auto cpg = torch::empty({blocks * 2048}, opts);
+  auto kfn = tokenize_v6_kernel<2048, true, 4096, 0>;
+  if (mode == 1) kfn = tokenize_v6_kernel<2048, true, 4096, 1>;
+  else if (mode == 2) kfn = tokenize_v6_kernel<2048, true, 4096, 2>;
+  else if (mode == 3) kfn = tokenize_v6_kernel<2048, true, 4096, 3>;
+  hipStream_t st = cur_stream();
+  hipEvent_t e0, e1;
+  hipEventCreate(&e0);
+  hipEventCreate(&e1);
+  auto launch = [&]() {
+    hipLaunchKernelGGL(kfn, dim3(blocks), dim3(kBlock), 0, st,
+                       text.data_ptr<u8>(), n, (u64)0, u64p(tkeys),
+                       tvals.data_ptr<i64>(), u64p(texm), (u64)(tcap - 1),
+                       u64p(oh), u64p(op),
+                       reinterpret_cast<unsigned long long*>(ctr.data_ptr<i64>()),
+                       cap,
+                       reinterpret_cast<unsigned long long*>(nw.data_ptr<i64>()),
+                       u64p(cpg));
+  };
+  launch();  // warm
+  hipEventRecord(e0, st);
+  for (long i = 0; i < iters; ++i) {
+    ctr.zero_();
+    launch();
+  }
+  hipEventRecord(e1, st);
+  hipEventSynchronize(e1);
+  float ms = 0;
+  hipEventElapsedTime(&ms, e0, e1);
+  hipEventDestroy(e0);
+  hipEventDestroy(e1);
+  return (double)ms / iters;
+}
+
 // diagnosis only
 double tok_ablate(torch::Tensor text, long mode, long iters) {
   long n = text.numel();
@@ -474,6 +522,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("tokenize_cache_spill", &tokenize_cache_spill,
         "tokenize; LDS cache counts the head, misses spill");
   m.def("tok_ablate", &tok_ablate, "ablation timing (diagnosis)");
+  m.def("tok6_ablate", &tok6_ablate, "v6 ablation timing");
   m.def("bucket_count", &bucket_count,
         "LDS count of bucket-partitioned (hash,pos)");
   m.def("hash_insert_count", &hash_insert_count);
