@@ -679,10 +679,12 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
     __syncthreads();
     for (int wnd = 0; wnd < TILE_N / 4096; ++wnd) {
     long my0 = ((long)tid + (long)wnd * 256) * TOK_BYTES;
-    if (my0 >= avail) break;
+    // whole wave stays in the loop (the spill wave-scan below needs every
+    // lane present); out-of-range lanes just contribute ns = 0
+    bool wactive = my0 < avail;
     // 32 bytes in registers: my window + 16B lookahead (halo-staged)
-    const uint4 va = *(const uint4*)&tile[my0];
-    const uint4 vb = *(const uint4*)&tile[my0 + 16];
+    const uint4 va = wactive ? *(const uint4*)&tile[my0] : uint4{0, 0, 0, 0};
+    const uint4 vb = wactive ? *(const uint4*)&tile[my0 + 16] : uint4{0, 0, 0, 0};
     const u32 rs[8] = {va.x, va.y, va.z, va.w, vb.x, vb.y, vb.z, vb.w};
     u32 m32 = 0;
     #pragma unroll
@@ -696,12 +698,13 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
       m32 |= (lim32 <= 0) ? 0xFFFFFFFFu : ~((1u << lim32) - 1);
     u32 prevb = (base + my0 == 0)
                     ? 1u
-                    : (u32)is_ws(my0 ? tile[my0 - 1] : text[base - 1]);
-    u32 sm = ~m32 & ((m32 << 1) | prevb) & 0xFFFFu;
+                    : (wactive ? (u32)is_ws(my0 ? tile[my0 - 1]
+                                                : text[base - 1]) : 1u);
+    u32 sm = wactive ? (~m32 & ((m32 << 1) | prevb) & 0xFFFFu) : 0u;
     long limw = avail - my0;  // starts must be real data bytes
     if (limw < 16) sm &= (limw <= 0) ? 0u : ((1u << limw) - 1);
     my_words += __popc(sm);
-    if (MODE == 1) { my_words += m32; continue; }
+    if (MODE == 1) { my_words += m32; sm = 0; }
     const u64 q0 = (u64)va.x | ((u64)va.y << 32);
     const u64 q1 = (u64)va.z | ((u64)va.w << 32);
     const u64 q2 = (u64)vb.x | ((u64)vb.y << 32);
