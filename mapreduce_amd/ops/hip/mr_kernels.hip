@@ -807,11 +807,15 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
         wbase = atomicAdd(spill_counter, (unsigned long long)wave_total);
       wbase = __shfl(wbase, 0, WAVE);
       long o = (long)wbase + (long)(incl - my_ns);
-      for (int w = 0; w < ns; ++w)
-        if (o + w < spill_cap) {
-          out_hash[o + w] = sh_[w];
-          out_pos[o + w] = sp_[w];
-        }
+      if (MODE == 4) {  // ablation: everything but the stores
+        for (int w = 0; w < ns; ++w) my_words += sh_[w] + sp_[w] + (u64)o;
+      } else {
+        for (int w = 0; w < ns; ++w)
+          if (o + w < spill_cap) {
+            out_hash[o + w] = sh_[w];
+            out_pos[o + w] = sp_[w];
+          }
+      }
     }
     }  // wnd
   }

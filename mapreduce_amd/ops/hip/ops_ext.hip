@@ -218,6 +218,7 @@ double tok6_ablate(torch::Tensor text, long mode, long iters) {
   if (mode == 1) kfn = tokenize_v6_kernel<2048, true, 4096, 1>;
   else if (mode == 2) kfn = tokenize_v6_kernel<2048, true, 4096, 2>;
   else if (mode == 3) kfn = tokenize_v6_kernel<2048, true, 4096, 3>;
+  else if (mode == 4) kfn = tokenize_v6_kernel<2048, true, 4096, 4>;
   hipStream_t st = cur_stream();
   hipEvent_t e0, e1;
   hipEventCreate(&e0);
