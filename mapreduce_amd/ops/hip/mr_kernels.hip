@@ -709,9 +709,6 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
     const u64 q1 = (u64)va.z | ((u64)va.w << 32);
     const u64 q2 = (u64)vb.x | ((u64)vb.y << 32);
     const u64 q3 = (u64)vb.z | ((u64)vb.w << 32);
-    u64 sh_[8];
-    u64 sp_[8];
-    int ns = 0;
     while (sm) {
       int s = __ffs(sm) - 1;
       sm &= sm - 1;
@@ -782,39 +779,34 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
         }
         slot = (slot + 1) & (CACHE_N - 1);
       }
-      if (!done) {
-        sh_[ns] = k;
-        sp_[ns] = p;
-        ++ns;
+      // per-iteration wave-aggregated spill: ballot over the lanes
+      // active at this trip of the word loop, ONE atomic per group,
+      // direct stores — no runtime-indexed register buffers (rule 20:
+      // divergent-index v_movrel waterfalls measured 2.7 ms of a 3.2 ms
+      // kernel)
+      bool miss = !done;
+      if (MODE == 3) {
+        my_words += miss;
+        miss = false;
       }
-    }
-    if (MODE == 3) { my_words += ns; ns = 0; }
-    // wave-aggregated spill reservation: ONE global atomic per wave per
-    // tile instead of one per thread — the per-thread same-counter
-    // atomicAdd measured 2.74 ms of a 3.22 ms kernel (ablation mode 3
-    // vs full); writes land wave-contiguous
-    {
-      u32 my_ns = (u32)ns;
-      u32 incl = my_ns;
-      #pragma unroll
-      for (int off = 1; off < WAVE; off <<= 1) {
-        u32 x = __shfl_up(incl, off, WAVE);
-        if ((threadIdx.x & (WAVE - 1)) >= off) incl += x;
-      }
-      u32 wave_total = __shfl(incl, WAVE - 1, WAVE);
-      unsigned long long wbase = 0;
-      if ((threadIdx.x & (WAVE - 1)) == 0 && wave_total)
-        wbase = atomicAdd(spill_counter, (unsigned long long)wave_total);
-      wbase = __shfl(wbase, 0, WAVE);
-      long o = (long)wbase + (long)(incl - my_ns);
-      if (MODE == 4) {  // ablation: everything but the stores
-        for (int w = 0; w < ns; ++w) my_words += sh_[w] + sp_[w] + (u64)o;
-      } else {
-        for (int w = 0; w < ns; ++w)
-          if (o + w < spill_cap) {
-            out_hash[o + w] = sh_[w];
-            out_pos[o + w] = sp_[w];
+      u64 mm = __ballot(miss);
+      if (mm) {
+        int lane = threadIdx.x & (WAVE - 1);
+        int leader = __ffsll((unsigned long long)mm) - 1;
+        unsigned long long o = 0;
+        if (lane == leader)
+          o = atomicAdd(spill_counter,
+                        (unsigned long long)__popcll(mm));
+        o = __shfl(o, leader, WAVE);
+        if (miss) {
+          long idx = (long)o + __popcll(mm & (((u64)1 << lane) - 1));
+          if (MODE == 4) {
+            my_words += (u64)idx + k + p;
+          } else if (idx < spill_cap) {
+            out_hash[idx] = k;
+            out_pos[idx] = p;
           }
+        }
       }
     }
     }  // wnd
