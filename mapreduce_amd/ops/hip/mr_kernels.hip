@@ -709,7 +709,18 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
     const u64 q1 = (u64)va.z | ((u64)va.w << 32);
     const u64 q2 = (u64)vb.x | ((u64)vb.y << 32);
     const u64 q3 = (u64)vb.z | ((u64)vb.w << 32);
-    while (sm) {
+    // fully-unrolled word loop (a 16-byte window holds <= 8 words):
+    // compile-time slot indices keep the miss buffers in REGISTERS —
+    // runtime-indexed arrays here lower to divergent v_movrel waterfalls
+    // (measured 2.7 ms of a 3.2 ms kernel, ablation modes 3 vs 4), and
+    // per-word ballot aggregation is worse still (12.8 ms: divergent
+    // ballots + atomics)
+    u64 sh_[8];
+    u64 sp_[8];
+    u32 miss_mask = 0;
+    #pragma unroll
+    for (int wi = 0; wi < 8; ++wi) {
+      if (!sm) break;
       int s = __ffs(sm) - 1;
       sm &= sm - 1;
       u32 t = m32 >> s;
@@ -722,7 +733,7 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
         // last window masks to the remaining bytes)
         for (u32 c = 0; c < len; c += 8) {
           int j0 = s + (int)c;
-          int qi = j0 >> 3;  // 0..4 (window may straddle past byte 31)
+          int qi = j0 >> 3;
           u64 lo = (qi & 2) ? ((qi & 1) ? q3 : q2) : ((qi & 1) ? q1 : q0);
           u64 hi = (qi >= 3) ? 0 : ((qi & 2) ? q3 : ((qi & 1) ? q2 : q1));
           int sh2 = 8 * (j0 & 7);
@@ -756,7 +767,7 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
       if (MODE == 2) { my_words += h; continue; }
       u64 k = remap_key(h);
       u64 p = ((pos_base + (u64)(base + my0 + s)) << 16) | (u64)len;
-      // LDS cache insert (v4 path)
+      // LDS cache insert
       u32 slot = (u32)((k ^ (k >> 32)) & (CACHE_N - 1));
       bool done = false;
       for (int pr = 0; pr < TOK_PROBE; ++pr) {
@@ -779,33 +790,43 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
         }
         slot = (slot + 1) & (CACHE_N - 1);
       }
-      // per-iteration wave-aggregated spill: ballot over the lanes
-      // active at this trip of the word loop, ONE atomic per group,
-      // direct stores — no runtime-indexed register buffers (rule 20:
-      // divergent-index v_movrel waterfalls measured 2.7 ms of a 3.2 ms
-      // kernel)
-      bool miss = !done;
-      if (MODE == 3) {
-        my_words += miss;
-        miss = false;
+      if (!done) {
+        if (MODE == 3) {
+          my_words += 1;
+        } else {
+          miss_mask |= 1u << wi;  // wi is compile-time: register slot
+          sh_[wi] = k;
+          sp_[wi] = p;
+        }
       }
-      u64 mm = __ballot(miss);
-      if (mm) {
-        int lane = threadIdx.x & (WAVE - 1);
-        int leader = __ffsll((unsigned long long)mm) - 1;
-        unsigned long long o = 0;
-        if (lane == leader)
-          o = atomicAdd(spill_counter,
-                        (unsigned long long)__popcll(mm));
-        o = __shfl(o, leader, WAVE);
-        if (miss) {
-          long idx = (long)o + __popcll(mm & (((u64)1 << lane) - 1));
+    }
+    // window-level spill reservation: every lane of the wave is present
+    // here (wactive design above) — wave-prefix the miss counts, one
+    // global atomic per wave per window, unrolled masked stores
+    {
+      u32 my_ns = (u32)__popc(miss_mask);
+      u32 incl = my_ns;
+      #pragma unroll
+      for (int off = 1; off < WAVE; off <<= 1) {
+        u32 x = __shfl_up(incl, off, WAVE);
+        if ((threadIdx.x & (WAVE - 1)) >= off) incl += x;
+      }
+      u32 wave_total = __shfl(incl, WAVE - 1, WAVE);
+      unsigned long long wbase = 0;
+      if ((threadIdx.x & (WAVE - 1)) == 0 && wave_total)
+        wbase = atomicAdd(spill_counter, (unsigned long long)wave_total);
+      wbase = __shfl(wbase, 0, WAVE);
+      long o = (long)wbase + (long)(incl - my_ns);
+      #pragma unroll
+      for (int wi = 0; wi < 8; ++wi) {
+        if (miss_mask & (1u << wi)) {
           if (MODE == 4) {
-            my_words += (u64)idx + k + p;
-          } else if (idx < spill_cap) {
-            out_hash[idx] = k;
-            out_pos[idx] = p;
+            my_words += sh_[wi] + sp_[wi] + (u64)o;
+          } else if (o < spill_cap) {
+            out_hash[o] = sh_[wi];
+            out_pos[o] = sp_[wi];
           }
+          ++o;
         }
       }
     }
