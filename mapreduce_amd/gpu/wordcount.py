@@ -125,7 +125,9 @@ class WordCountJob:
         self.reset()
         self._text = text
         if self.mode == "streaming":
-            cap = text.numel() // 2 + 16
+            # + slack for the wave-chunked spill allocator's padded chunk
+            # tails (<= 2048 blocks x 4 waves x 512-entry chunks)
+            cap = text.numel() // 2 + 16 + 2048 * 4 * 512
             opts = dict(dtype=torch.int64, device=self.device)
             self._spill_h = torch.empty(cap, **opts)
             self._spill_p = torch.empty(cap, **opts)
