@@ -108,10 +108,15 @@ class InvertedIndexJob:
         starts = torch.tensor([s for s, _ in splits] + [splits[-1][1]],
                               device=dev, dtype=torch.int64)
         if dev.type == "cuda":
-            cap = text.numel() // 2 + 16
-            h, p, c = ops.ext().tokenize_spill(text, 0, cap)
+            # spill-all tokenizer (wave-chunked allocator); chunk-tail
+            # padding carries HT_EMPTY keys (= -1 as int64) — filter once
+            cap = text.numel() // 2 + 16 + 2048 * 4 * 512
+            h, p, c, nw = ops.ext().tokenize_spill_v2(text, 0, cap)
             n = int(c.item())
             h, p = h[:n], p[:n]
+            real = h != -1
+            h = h[real]
+            p = p[real]
         else:
             h, p, n = ops.tokenize_words(text)
         byte_off = p >> 16

@@ -641,7 +641,9 @@ __global__ __launch_bounds__(256) void tokenize_v5_kernel(
 // ---------------------------------------------------------------------------
 
 // MODE (ablation, §5.4 rule 8): 0=full, 1=stage+classify, 2=+hash, 3=+cache (no spill)
-template <int CACHE_N, bool GPOS, int TILE_N, int MODE = 0>
+template <int CACHE_N, bool GPOS, int TILE_N, int MODE = 0,
+          bool SPILL_ALL = false>  // SPILL_ALL: emit every word (no
+                                   // cache) — the inverted-index path
 __global__ __launch_bounds__(256) void tokenize_v6_kernel(
     const u8* __restrict__ text, long n, u64 pos_base,
     u64* __restrict__ tkeys, i64* __restrict__ tvals, u64* __restrict__ texm,
@@ -776,6 +778,12 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
       if (MODE == 2) { my_words += h; continue; }
       u64 k = remap_key(h);
       u64 p = ((pos_base + (u64)(base + my0 + s)) << 16) | (u64)len;
+      if (SPILL_ALL) {
+        miss_mask |= 1u << wi;
+        sh_[wi] = k;
+        sp_[wi] = p;
+        continue;
+      }
       // LDS cache insert
       u32 slot = (u32)((k ^ (k >> 32)) & (CACHE_N - 1));
       bool done = false;

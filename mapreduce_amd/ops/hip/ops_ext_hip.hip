@@ -96,6 +96,36 @@ void tokenize_count(torch::Tensor text, long pos_base, torch::Tensor tkeys,
                      reinterpret_cast<unsigned long long*>(nwords.data_ptr<i64>()));
 }
 
+// ---------------------------------------------------- K2 streaming (v2)
+// every word spilled via the wave-chunked allocator (no cache) — chunk
+// tails are HT_EMPTY-padded, so consumers must filter/skip them.  The
+// inverted-index tokenize path; replaces tokenize_spill's per-window
+// shared-counter atomics (cross-XCD contention, see tokenize_v6 note).
+std::vector<torch::Tensor> tokenize_spill_v2(torch::Tensor text,
+                                             long pos_base, long cap) {
+  TORCH_CHECK(text.is_cuda() && text.scalar_type() == torch::kUInt8 &&
+              text.is_contiguous(), "text must be contiguous u8 on GPU");
+  long n = text.numel();
+  auto opts = torch::TensorOptions().device(text.device()).dtype(torch::kInt64);
+  auto out_hash = torch::empty({cap}, opts);
+  auto out_pos = torch::empty({cap}, opts);
+  auto counter = torch::zeros({1}, opts);
+  auto nwords = torch::zeros({1}, opts);
+  if (n) {
+    auto dummy = torch::empty({16}, opts);  // table unused in SPILL_ALL
+    hipLaunchKernelGGL((tokenize_v6_kernel<16, false, 4096, 0, true>),
+                       dim3(grid_for(n, TOK_BYTES)), dim3(kBlock), 0,
+                       cur_stream(), text.data_ptr<u8>(), n, (u64)pos_base,
+                       u64p(dummy), dummy.data_ptr<i64>(), nullptr,
+                       (u64)15, u64p(out_hash), u64p(out_pos),
+                       reinterpret_cast<unsigned long long*>(counter.data_ptr<i64>()),
+                       cap,
+                       reinterpret_cast<unsigned long long*>(nwords.data_ptr<i64>()),
+                       nullptr);
+  }
+  return {out_hash, out_pos, counter, nwords};
+}
+
 // ------------------------------------------------------------- K2 streaming
 std::vector<torch::Tensor> tokenize_spill(torch::Tensor text, long pos_base,
                                           long cap) {
@@ -519,6 +549,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("tokenize", &tokenize, "tokenize text -> (hash, pos, count)");
   m.def("tokenize_count", &tokenize_count,
         "fused tokenize + hash-table count");
+  m.def("tokenize_spill_v2", &tokenize_spill_v2,
+        "spill-all tokenizer (chunk-padded; filter HT_EMPTY)");
   m.def("tokenize_spill", &tokenize_spill,
         "tokenize -> compact (hash,pos) arrays");
   m.def("tokenize_cache_spill", &tokenize_cache_spill,
