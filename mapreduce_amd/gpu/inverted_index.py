@@ -57,11 +57,13 @@ class InvertedIndexJob:
         self.rank, self.world = dx.world_info(group)
         self.doc_base = doc_base  # global id of this rank's first doc
 
-    def _sort_by_doc_then_hash(self, h, d, p):
-        # LSD composite: stable sorts, least-significant key first
+    def _sort_by_doc_then_hash(self, h, d, p, max_doc: int = 1 << 30):
+        # LSD composite: stable sorts, least-significant key first; doc
+        # ids are small, so the doc pass sorts only the bits that exist
         d1, h1, p1 = (d, h, p)
         if h.is_cuda:
-            d1, h1, p1 = ops.sort_by_key(d, h, p, bits=32)
+            doc_bits = max(8, int(max_doc).bit_length())
+            d1, h1, p1 = ops.sort_by_key(d, h, p, bits=doc_bits)
             h2, d2, p2 = ops.sort_by_key(h1, d1, p1, bits=64)
         else:
             d1, h1, p1 = ops.sort_by_key(d, h, p, bits=64)
@@ -122,7 +124,8 @@ class InvertedIndexJob:
         byte_off = p >> 16
         d = torch.searchsorted(starts, byte_off, right=True) - 1
         d = d + self.doc_base
-        h, d, p = self._sort_by_doc_then_hash(h, d, p)
+        max_doc = self.doc_base + len(splits) + 1
+        h, d, p = self._sort_by_doc_then_hash(h, d, p, max_doc)
         uh, ud, tf, up = self._segment_pairs(h, d, p)
 
         blob_src = text
@@ -159,7 +162,8 @@ class InvertedIndexJob:
             # composite re-sort (carrying pos + tf via an index payload) +
             # merge duplicate (hash, doc) pairs arriving from several ranks
             idx = torch.arange(rh.numel(), device=dev, dtype=torch.int64)
-            hs, ds, perm = self._sort_by_doc_then_hash(rh, rd, idx)
+            hs, ds, perm = self._sort_by_doc_then_hash(
+                rh, rd, idx, (self.doc_base + len(splits) + 1) * self.world)
             ps = rp.index_select(0, perm)
             tfs = rtf.index_select(0, perm)
             if hs.is_cuda and hs.numel():
