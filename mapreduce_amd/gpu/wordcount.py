@@ -159,7 +159,9 @@ class WordCountJob:
                 bucket_off = torch.zeros(257, dtype=torch.int64,
                                          device=self.device)
                 torch.cumsum(totals, 0, out=bucket_off[1:])
-                ops.ext().bucket_count(hk, pv, bucket_off, 256, 8,
+                import os
+                slices = int(os.environ.get("MR_BUCKET_SLICES", "8"))
+                ops.ext().bucket_count(hk, pv, bucket_off, 256, slices,
                                        self.table.tkeys, self.table.tvals,
                                        self.table.texm)
             self._spill_h = self._spill_p = None
