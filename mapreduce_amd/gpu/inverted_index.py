@@ -115,6 +115,8 @@ class InvertedIndexJob:
             cap = text.numel() // 2 + 16 + 2048 * 4 * 512
             h, p, c, nw = ops.ext().tokenize_spill_v2(text, 0, cap)
             n = int(c.item())
+            if n > cap:
+                raise RuntimeError(f"spill overflow: {n} reserved > {cap}")
             h, p = h[:n], p[:n]
             real = h != -1
             h = h[real]

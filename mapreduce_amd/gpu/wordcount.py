@@ -152,6 +152,13 @@ class WordCountJob:
         if self.mode == "streaming":
             nspill = int(self._spill_c.item())
             n = int(self._nwords.item())
+            if nspill > self._spill_cap:
+                # chunk reservations past the cap were dropped in-kernel;
+                # unwritten tail slots would read as garbage keys — fail
+                # loudly instead (sizing bug: grow the allocator slack)
+                raise RuntimeError(
+                    f"spill overflow: {nspill} reserved > cap "
+                    f"{self._spill_cap}")
             if nspill:
                 h = self._spill_h[:nspill]
                 p = self._spill_p[:nspill]

@@ -170,7 +170,12 @@ class StoreCoordinator(Coordinator):
     def get_doc(self, key: str):
         if not self._store.check([self._k(key)]):
             return (None, None)
-        raw = bytes(self._store.get(self._k(key)))
+        try:
+            raw = bytes(self._store.get(self._k(key)))
+        except Exception:
+            # deleted between check and get (e.g. server teardown racing a
+            # worker poll) — treat as absent
+            return (None, None)
         return _dec(raw), raw
 
     def cas_doc(self, key: str, expected_raw: Optional[bytes], doc: dict) -> bool:
