@@ -241,6 +241,25 @@ def test_streaming_wordcount_mode_vs_counter(dev):
     assert dict(res2.to_host()) == got
 
 
+def test_cluster_runner_dynamic_gpu(dev):
+    """Control-plane dynamic claims driving GPU map jobs (per-job CAS,
+    WRITTEN transitions) — single rank on hardware."""
+    import collections
+    from mapreduce_amd.gpu.corpus import make_corpus
+    from mapreduce_amd.gpu.runner import GpuClusterRunner
+    from mapreduce_amd.gpu.wordcount import WordCountJob
+    c = make_corpus(dev, nwords=100_000, nsplits=6, vocab_size=2000, seed=3)
+    job = WordCountJob(dev, vocab_estimate=4000)
+    runner = GpuClusterRunner(job, claim_mode="dynamic")
+    res = runner.run(c.text, c.splits())
+    assert res.nwords == 100_000
+    got = dict(res.to_host())
+    exp = collections.Counter(bytes(c.text.cpu().numpy().tobytes()).split())
+    assert got == dict(exp)
+    st = runner.job_stats()
+    assert st == {"jobs": 6, "written": 6, "broken": 0}
+
+
 def test_streamed_file_wordcount_gpu(dev, tmp_path):
     """K8 path on hardware: pinned staging + side-stream H2D chunks feed
     the engine; counts equal the file oracle; phase tracing populates."""
