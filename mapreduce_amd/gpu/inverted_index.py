@@ -50,6 +50,25 @@ class InvertedIndexResult:
         return out
 
 
+_SM64_C0 = 0x9E3779B97F4A7C15 - (1 << 64)
+_SM64_C1 = 0xBF58476D1CE4E5B9 - (1 << 64)
+_SM64_C2 = 0x94D049BB133111EB - (1 << 64)
+
+
+def _lsr(x: torch.Tensor, k: int) -> torch.Tensor:
+    """Logical shift right on int64 bit patterns (torch >> is arithmetic)."""
+    return (x >> k) & ((1 << (64 - k)) - 1)
+
+
+def splitmix64_t(x: torch.Tensor) -> torch.Tensor:
+    """SplitMix64 on int64 bit patterns (wrapping mul) — must match
+    utils.tuple.splitmix64; unit-tested against it."""
+    z = x + _SM64_C0
+    z = (z ^ _lsr(z, 30)) * _SM64_C1
+    z = (z ^ _lsr(z, 27)) * _SM64_C2
+    return z ^ _lsr(z, 31)
+
+
 class InvertedIndexJob:
     def __init__(self, device, group=None, doc_base: int = 0):
         self.device = torch.device(device)
@@ -109,6 +128,8 @@ class InvertedIndexJob:
             splits = [(0, int(text.numel()))]
         starts = torch.tensor([s for s, _ in splits] + [splits[-1][1]],
                               device=dev, dtype=torch.int64)
+        max_doc = self.doc_base + len(splits) + 1
+        doc_bits = max(8, int(max_doc).bit_length())
         if dev.type == "cuda":
             # spill-all tokenizer (wave-chunked allocator); chunk-tail
             # padding carries HT_EMPTY keys (= -1 as int64) — filter once
@@ -121,14 +142,37 @@ class InvertedIndexJob:
             real = h != -1
             h = h[real]
             p = p[real]
+            # ---- aggregate BEFORE sorting: composite (word, doc) keys
+            # through the bucketize + LDS-count machinery collapse the
+            # token stream to unique postings (~7x fewer elements for the
+            # radix passes; sorting the raw stream measured ~20 ms of a
+            # 26 ms job).  doc is recoverable from the exemplar position,
+            # so wordhash = k2 ^ splitmix64(doc) reverses the composite.
+            d = torch.searchsorted(starts, p >> 16, right=True) - 1
+            d = d + self.doc_base
+            k2 = h ^ splitmix64_t(d)
+            k2 = torch.where(k2 == -1, torch.full_like(k2, -2), k2)
+            hk, pv, totals = ops.ext().radix_pass(k2, p, 56)
+            bucket_off = torch.zeros(257, dtype=torch.int64, device=dev)
+            torch.cumsum(totals, 0, out=bucket_off[1:])
+            table = ops.make_table(max(1 << 16, k2.numel() // 4), dev)
+            ops.ext().bucket_count(hk, pv, bucket_off, 256, 32,
+                                   table.tkeys, table.tvals, table.texm)
+            uk2, tf, upos = table.extract()
+            ud = torch.searchsorted(starts, upos >> 16, right=True) - 1
+            ud = ud + self.doc_base
+            uh = uk2 ^ splitmix64_t(ud)
+            # output order: doc within word, words by hash
+            d1, h1, tf1, p1 = ops.sort_by_key(ud, uh, tf, upos,
+                                              bits=doc_bits)
+            uh, ud, tf, up = ops.sort_by_key(h1, d1, tf1, p1, bits=64)
         else:
             h, p, n = ops.tokenize_words(text)
-        byte_off = p >> 16
-        d = torch.searchsorted(starts, byte_off, right=True) - 1
-        d = d + self.doc_base
-        max_doc = self.doc_base + len(splits) + 1
-        h, d, p = self._sort_by_doc_then_hash(h, d, p, max_doc)
-        uh, ud, tf, up = self._segment_pairs(h, d, p)
+            byte_off = p >> 16
+            d = torch.searchsorted(starts, byte_off, right=True) - 1
+            d = d + self.doc_base
+            h, d, p = self._sort_by_doc_then_hash(h, d, p, max_doc)
+            uh, ud, tf, up = self._segment_pairs(h, d, p)
 
         blob_src = text
         if self.world > 1:

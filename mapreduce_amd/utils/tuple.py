@@ -83,6 +83,16 @@ def wordhash64(data: Any) -> int:
     return ((h ^ len(b)) * FNV64_PRIME) & _MASK64
 
 
+def splitmix64(x: int) -> int:
+    """SplitMix64 finalizer — the doc-id mixer for composite
+    (word, doc) keys in the GPU inverted index (must match the torch
+    implementation in gpu/inverted_index.py)."""
+    z = (x + 0x9E3779B97F4A7C15) & _MASK64
+    z = ((z ^ (z >> 30)) * 0xBF58476D1CE4E5B9) & _MASK64
+    z = ((z ^ (z >> 27)) * 0x94D049BB133111EB) & _MASK64
+    return (z ^ (z >> 31)) & _MASK64
+
+
 def jenkins_oaat(data: Any) -> int:
     """Jenkins one-at-a-time 32-bit hash (tuple.lua:121-140)."""
     h = 0

@@ -97,6 +97,22 @@ def test_sort_key_nested_after_tuple_module_import():
     assert sort_key(it) == sort_key((1, (2, 3)))
 
 
+def test_splitmix64_torch_matches_python():
+    import torch
+    from mapreduce_amd.gpu.inverted_index import splitmix64_t
+    from mapreduce_amd.utils.tuple import splitmix64
+    import numpy as np
+
+    rng = np.random.default_rng(8)
+    xs = rng.integers(0, 2 ** 63, size=1000, dtype=np.uint64)
+    xs[0] = 0
+    xs[1] = 2 ** 63 - 1
+    t = torch.from_numpy(xs.view(np.int64))
+    got = splitmix64_t(t).numpy().view(np.uint64)
+    exp = np.array([splitmix64(int(x)) for x in xs], dtype=np.uint64)
+    assert np.array_equal(got, exp)
+
+
 def test_record_roundtrip():
     buf = io.BytesIO()
     rows = [("a", [1]), ((1, 2), [1, 2, 3]), (5, ["x"])]
