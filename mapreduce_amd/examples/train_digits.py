@@ -19,7 +19,6 @@ from __future__ import annotations
 
 import base64
 import io
-import pickle
 
 import torch
 

@@ -9,7 +9,6 @@ Generation happens once per run on the GPU (not timed)."""
 
 from __future__ import annotations
 
-import math
 from dataclasses import dataclass
 from typing import List
 

@@ -10,7 +10,7 @@ profiles/, the op is bandwidth-bound so VALU wins)."""
 
 from __future__ import annotations
 
-from typing import Dict, List
+from typing import Dict
 
 import torch
 import torch.distributed as dist

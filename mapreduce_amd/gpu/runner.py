@@ -22,7 +22,6 @@ Two claim modes:
 
 from __future__ import annotations
 
-import time
 from typing import List, Optional, Tuple
 
 import torch

@@ -22,7 +22,6 @@ precondition (job.lua:264-274)."""
 
 from __future__ import annotations
 
-import time
 from dataclasses import dataclass
 from typing import List, Optional, Tuple
 

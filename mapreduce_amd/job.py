@@ -19,8 +19,7 @@ from __future__ import annotations
 import importlib
 import re
 import time
-import traceback
-from typing import Any, Callable, Dict, List, Optional
+from typing import Any, Callable, Dict, List
 
 from . import fs as fsmod
 from .parallel.coord import Coordinator

@@ -7,7 +7,7 @@ are never silently used (ops.require_gpu_ext)."""
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Optional
 
 import numpy as np
 import torch

@@ -15,8 +15,7 @@ relaxes to any claimable job.
 
 from __future__ import annotations
 
-import time
-from typing import Any, Dict, List, Optional, Tuple
+from typing import Any, List, Optional, Tuple
 
 from .parallel.coord import Coordinator
 from .utils import (MAX_IDLE_COUNT, MAX_JOB_RETRIES, STATUS, TASK_STATUS,

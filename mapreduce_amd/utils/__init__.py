@@ -19,11 +19,10 @@ Reference parity map:
 from __future__ import annotations
 
 import io
-import math
 import pickle
 import struct
 import time
-from typing import Any, Callable, Iterable, Iterator, List, Tuple
+from typing import Any, Iterable, Iterator, List, Tuple
 
 from .heap import Heap
 

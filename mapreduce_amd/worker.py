@@ -23,7 +23,7 @@ from typing import Optional
 from .job import FnSet, Job
 from .parallel.coord import Coordinator, connect
 from .task import Task
-from .utils import DEFAULT_SLEEP, MAX_WORKER_RETRIES, TASK_STATUS
+from .utils import DEFAULT_SLEEP, MAX_WORKER_RETRIES
 
 
 class Worker:
