@@ -1033,13 +1033,14 @@ __global__ __launch_bounds__(256) void bucket_count_kernel(
     ccnt[s] = 0;
   }
   __syncthreads();
-  // two elements in flight per thread: the insert chain is LDS-latency
-  // bound (PMC: 8% VALU-busy), so overlapping two independent probes
-  // hides half the latency
-  auto insert_one = [&](u64 k, u64 p, u64 cur0, u32 slot) {
+  for (long i = s0 + threadIdx.x; i < s1; i += blockDim.x) {
+    u64 k = hashes[i];
+    if (k == HT_EMPTY) continue;  // spill-chunk padding (tokenize_v6)
+    u64 p = pos[i];
+    u32 slot = (u32)((k ^ (k >> 17)) & (BKT_SLOTS - 1));
     bool done = false;
-    u64 cur = cur0;
     for (int pr = 0; pr < 64; ++pr) {
+      u64 cur = ckeys[slot];
       if (cur == k) {
         atomicAdd(&ccnt[slot], 1u);
         done = true;
@@ -1055,33 +1056,11 @@ __global__ __launch_bounds__(256) void bucket_count_kernel(
           done = true;
           break;
         }
-        cur = ckeys[slot];
-        continue;
       }
       slot = (slot + 1) & (BKT_SLOTS - 1);
-      cur = ckeys[slot];
     }
     if (!done)  // pathological bucket: spill straight to the global table
       ht_add(k, p, 1, tkeys, tvals, texm, cap_mask);
-  };
-  long i = s0 + threadIdx.x;
-  for (; i + blockDim.x < s1; i += 2 * blockDim.x) {
-    u64 ka = hashes[i];
-    u64 kb = hashes[i + blockDim.x];
-    u64 pa = pos[i];
-    u64 pb = pos[i + blockDim.x];
-    u32 sa = (u32)((ka ^ (ka >> 17)) & (BKT_SLOTS - 1));
-    u32 sb = (u32)((kb ^ (kb >> 17)) & (BKT_SLOTS - 1));
-    u64 ca = ckeys[sa];  // both first probes issue before either resolves
-    u64 cb = ckeys[sb];
-    if (ka != HT_EMPTY) insert_one(ka, pa, ca, sa);
-    if (kb != HT_EMPTY) insert_one(kb, pb, cb, sb);
-  }
-  for (; i < s1; i += blockDim.x) {
-    u64 k = hashes[i];
-    if (k == HT_EMPTY) continue;  // spill-chunk padding (tokenize_v6)
-    u32 slot = (u32)((k ^ (k >> 17)) & (BKT_SLOTS - 1));
-    insert_one(k, pos[i], ckeys[slot], slot);
   }
   __syncthreads();
   for (int s = threadIdx.x; s < BKT_SLOTS; s += blockDim.x)
