@@ -94,8 +94,14 @@ def make_corpus(device, nwords: int = EUROPARL_WORDS,
 
     g = torch.Generator(device=dev)
     g.manual_seed(seed)
-    u = torch.rand(nwords, generator=g, device=dev, dtype=torch.float64)
-    ids = torch.searchsorted(cdf, u).clamp_(max=vocab_size - 1)
+    # chunked sampling: ATen's searchsorted rejects >2^31-element launches
+    ids = torch.empty(nwords, dtype=torch.int64, device=dev)
+    CH = 1 << 29
+    for off in range(0, nwords, CH):
+        m = min(CH, nwords - off)
+        u = torch.rand(m, generator=g, device=dev, dtype=torch.float64)
+        torch.searchsorted(cdf, u, out=ids[off:off + m])
+    ids.clamp_(max=vocab_size - 1)
 
     lens_t = torch.from_numpy(lens).to(dev)
     voff_t = torch.from_numpy(voff.astype(np.int64)).to(dev)
