@@ -1,0 +1,62 @@
+"""CLI end-to-end: the execute_server / execute_worker entry points with
+real processes (reference execute_server.lua:25-62 / execute_worker.lua
+parity, including the "nil" sentinel and dot-form module names)."""
+
+import collections
+import json
+import os
+import socket
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+@pytest.mark.timeout(180)
+def test_server_worker_cli_end_to_end(tmp_path):
+    files = []
+    for i in range(3):
+        p = tmp_path / f"f{i}.txt"
+        p.write_text("red green blue red\nred green\n" * (10 * (i + 1)))
+        files.append(str(p))
+    out = tmp_path / "result.txt"
+    cnn = f"tcp://127.0.0.1:{free_port()}"
+    env = dict(os.environ, PYTHONPATH=REPO)
+    m = "mapreduce_amd.examples.wordcount"
+    worker = subprocess.Popen(
+        [sys.executable, "-m", "mapreduce_amd.execute_worker", cnn, "cliwc",
+         "--max-iter", "1000000", "--max-tasks", "1000000"],
+        env=env, cwd=REPO)
+    try:
+        server = subprocess.run(
+            [sys.executable, "-m", "mapreduce_amd.execute_server", cnn,
+             "cliwc", m, m, m, m, m, "nil",
+             "--storage", f"shared:{tmp_path}/shuffle",
+             "--init-args", json.dumps({"files": files, "out": str(out)}),
+             "--sleep", "0.1"],
+            env=env, cwd=REPO, capture_output=True, text=True, timeout=120)
+        assert server.returncode == 0, server.stderr[-2000:]
+        got = {}
+        for line in out.read_text().splitlines():
+            c, w = line.split("\t")
+            got[w] = int(c)
+        exp = collections.Counter()
+        for f in files:
+            exp.update(open(f).read().split())
+        assert got == dict(exp)
+    finally:
+        worker.terminate()
+        try:
+            worker.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            worker.kill()
