@@ -152,20 +152,11 @@ class InvertedIndexJob:
             d = d + self.doc_base
             k2 = h ^ splitmix64_t(d)
             k2 = torch.where(k2 == -1, torch.full_like(k2, -2), k2)
-            # TWO bucketize passes -> 65536 buckets: the composite key
-            # space has ~n/7 distinct entries, so 256 buckets leave ~29k
-            # distinct per bucket (>> the 2048 LDS slots — most inserts
-            # fell through to latency-bound global-table probes); at 16
-            # bucket bits every bucket's distinct set fits in LDS
-            hk, pv, _ = ops.ext().radix_pass(k2, p, 48)
-            hk, pv, _ = ops.ext().radix_pass(hk, pv, 56)
-            b16 = _lsr(hk, 48)
-            counts16 = torch.bincount(b16, minlength=1 << 16)
-            bucket_off = torch.zeros((1 << 16) + 1, dtype=torch.int64,
-                                     device=dev)
-            torch.cumsum(counts16, 0, out=bucket_off[1:])
+            hk, pv, totals = ops.ext().radix_pass(k2, p, 56)
+            bucket_off = torch.zeros(257, dtype=torch.int64, device=dev)
+            torch.cumsum(totals, 0, out=bucket_off[1:])
             table = ops.make_table(max(1 << 16, k2.numel() // 4), dev)
-            ops.ext().bucket_count(hk, pv, bucket_off, 1 << 16, 1,
+            ops.ext().bucket_count(hk, pv, bucket_off, 256, 32,
                                    table.tkeys, table.tvals, table.texm)
             uk2, tf, upos = table.extract()
             ud = torch.searchsorted(starts, upos >> 16, right=True) - 1
