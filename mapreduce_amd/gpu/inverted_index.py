@@ -155,7 +155,7 @@ class InvertedIndexJob:
             hk, pv, totals = ops.ext().radix_pass(k2, p, 56)
             bucket_off = torch.zeros(257, dtype=torch.int64, device=dev)
             torch.cumsum(totals, 0, out=bucket_off[1:])
-            table = ops.make_table(max(1 << 16, k2.numel() // 4), dev)
+            table = ops.make_table(max(1 << 16, k2.numel() // 6), dev)
             ops.ext().bucket_count(hk, pv, bucket_off, 256, 32,
                                    table.tkeys, table.tvals, table.texm)
             uk2, tf, upos = table.extract()

@@ -106,7 +106,22 @@ class HashTable:
         ext().hash_insert_sum_i64(keys, vals, self.tkeys, self.tvals, n)
 
     def extract(self):
-        """-> (keys, vals, pos) compacted, unsorted.  One sync for count."""
+        """-> (keys, vals, pos) compacted, unsorted.  One sync for count.
+
+        Large tables use the chunked-compaction kernel (the per-wave
+        shared-counter atomic serializes cross-XCD — measured 6.3 ms on a
+        2^25-slot scan) and mask out its HT_EMPTY chunk padding here;
+        small tables keep the padding-free path."""
+        if self.cap >= (1 << 22):
+            k, v, p, c = ext().hash_extract_v2(self.tkeys, self.tvals,
+                                               self.texm)
+            n = int(c.item())
+            k = k[:n]
+            real = k != -1
+            k = k[real]
+            v = v[:n][real]
+            p = p[:n][real] if p.numel() else p
+            return k, v, p
         k, v, p, c = ext().hash_extract(self.tkeys, self.tvals, self.texm)
         n = int(c.item())
         return k[:n], v[:n], (p[:n] if p.numel() else p)

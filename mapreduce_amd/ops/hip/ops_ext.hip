@@ -375,6 +375,27 @@ std::vector<torch::Tensor> hash_extract(torch::Tensor tkeys,
   return {okeys, ovals, opos, counter};
 }
 
+std::vector<torch::Tensor> hash_extract_v2(torch::Tensor tkeys,
+                                           torch::Tensor tvals,
+                                           torch::Tensor texm) {
+  long cap = tkeys.numel();
+  auto opts = tkeys.options();
+  long waves = (2048L * kBlock) / 64;
+  long ocap = cap + waves * 256;  // chunk-tail padding slack
+  auto okeys = torch::empty({ocap}, opts);
+  auto ovals = torch::empty({ocap}, opts);
+  bool exm = texm.numel() > 0;
+  auto opos = torch::empty({exm ? ocap : 0}, opts);
+  auto counter = torch::zeros({1}, opts);
+  hipLaunchKernelGGL(hash_extract_v2_kernel, dim3(2048), dim3(kBlock), 0,
+                     cur_stream(), u64cp(tkeys), tvals.data_ptr<i64>(),
+                     exm ? u64cp(texm) : nullptr, cap, u64p(okeys),
+                     ovals.data_ptr<i64>(), exm ? u64p(opos) : nullptr,
+                     reinterpret_cast<unsigned long long*>(counter.data_ptr<i64>()),
+                     ocap);
+  return {okeys, ovals, opos, counter};
+}
+
 // ----------------------------------------------------------------------- K5b
 torch::Tensor head_flags(torch::Tensor keys) {
   check_dev_i64(keys, "keys");
@@ -561,6 +582,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hash_insert_count", &hash_insert_count);
   m.def("hash_insert_sum_i64", &hash_insert_sum_i64);
   m.def("hash_extract", &hash_extract);
+  m.def("hash_extract_v2", &hash_extract_v2,
+        "chunked compaction (HT_EMPTY-padded; mask/trim downstream)");
   m.def("head_flags", &head_flags);
   m.def("seg_reduce_i64", &seg_reduce_i64);
   m.def("seg_first_u64", &seg_first_u64);

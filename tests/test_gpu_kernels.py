@@ -91,6 +91,24 @@ def test_hash_table_sum_vs_numpy(dev):
     assert got == exp
 
 
+def test_hash_table_big_capacity_chunked_extract(dev):
+    """Tables >= 2^22 slots use the chunked-compaction extract (HT_EMPTY
+    padding masked in the wrapper) — verify exact contents at scale."""
+    from mapreduce_amd import ops
+    n = 3_000_000
+    base = torch.arange(n, dtype=torch.int64, device=dev) * 2654435761
+    keys = torch.cat([base, base])  # every key exactly twice
+    ht = ops.HashTable(n, dev, exemplar=False)
+    assert ht.cap >= (1 << 22)
+    ht.insert_sum(keys, torch.ones_like(keys))
+    uk, uv, _ = ht.extract()
+    assert uk.numel() == n
+    assert bool((uv == 2).all())
+    sk, sv = ops.sort_pairs(uk, uv)
+    exp = np.sort(base.cpu().numpy().view(np.uint64))
+    assert np.array_equal(sk.cpu().numpy().view(np.uint64), exp)
+
+
 @pytest.mark.parametrize("n", [0, 1, 63, 64, 2048, 2049, 1_000_000])
 def test_radix_sort_keys(dev, n):
     from mapreduce_amd import ops
