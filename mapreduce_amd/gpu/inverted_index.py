@@ -134,24 +134,22 @@ class InvertedIndexJob:
             # spill-all tokenizer (wave-chunked allocator); chunk-tail
             # padding carries HT_EMPTY keys (= -1 as int64) — filter once
             cap = text.numel() // 2 + 16 + 2048 * 4 * 512
-            h, p, c, nw = ops.ext().tokenize_spill_v2(text, 0, cap)
+            # ---- aggregate BEFORE sorting: the tokenizer emits composite
+            # (word, doc) keys directly (doc looked up in-kernel from the
+            # split offsets; wordhash ^ splitmix64(doc)), which the
+            # bucketize + LDS-count machinery collapses to unique postings
+            # (~7x fewer elements for the radix passes; sorting the raw
+            # stream measured ~20 ms of a 26 ms job).  doc is recoverable
+            # from the exemplar position, so wordhash = k2 ^
+            # splitmix64(doc) reverses the composite.  HT_EMPTY chunk
+            # padding flows through — the radix pass groups it in bucket
+            # 255 and bucket_count skips it.
+            k2, p, c, nw = ops.ext().tokenize_spill_composite(
+                text, 0, cap, starts, self.doc_base)
             n = int(c.item())
             if n > cap:
                 raise RuntimeError(f"spill overflow: {n} reserved > {cap}")
-            h, p = h[:n], p[:n]
-            real = h != -1
-            h = h[real]
-            p = p[real]
-            # ---- aggregate BEFORE sorting: composite (word, doc) keys
-            # through the bucketize + LDS-count machinery collapse the
-            # token stream to unique postings (~7x fewer elements for the
-            # radix passes; sorting the raw stream measured ~20 ms of a
-            # 26 ms job).  doc is recoverable from the exemplar position,
-            # so wordhash = k2 ^ splitmix64(doc) reverses the composite.
-            d = torch.searchsorted(starts, p >> 16, right=True) - 1
-            d = d + self.doc_base
-            k2 = h ^ splitmix64_t(d)
-            k2 = torch.where(k2 == -1, torch.full_like(k2, -2), k2)
+            k2, p = k2[:n], p[:n]
             hk, pv, totals = ops.ext().radix_pass(k2, p, 56)
             bucket_off = torch.zeros(257, dtype=torch.int64, device=dev)
             torch.cumsum(totals, 0, out=bucket_off[1:])

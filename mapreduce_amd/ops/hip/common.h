@@ -36,6 +36,25 @@ DEV u32 partition_of(u64 h, u32 nparts) { return (u32)mulhi_u64(h, (u64)nparts);
 DEV u64 whash_chunk(u64 h, u64 chunk) { return (h ^ chunk) * FNV64_PRIME; }
 DEV u64 whash_fin(u64 h, u64 len) { return (h ^ len) * FNV64_PRIME; }
 
+// SplitMix64 finalizer (must match utils.tuple.splitmix64)
+DEV u64 splitmix64_dev(u64 x) {
+  u64 z = x + 0x9E3779B97F4A7C15ull;
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+  return z ^ (z >> 31);
+}
+
+// index of last element <= v in a sorted i64 array (upper_bound - 1)
+DEV int ub_minus1(const i64* a, int n, i64 v) {
+  int lo = 0, hi = n;  // first index with a[i] > v
+  while (lo < hi) {
+    int mid = (lo + hi) >> 1;
+    if (a[mid] <= v) lo = mid + 1;
+    else hi = mid;
+  }
+  return lo - 1;
+}
+
 DEV bool is_ws(u8 c) {
   // Python str.split() whitespace set: \t \n \v \f \r ' '
   return c == ' ' || (c >= 9 && c <= 13);

@@ -642,14 +642,19 @@ __global__ __launch_bounds__(256) void tokenize_v5_kernel(
 
 // MODE (ablation, §5.4 rule 8): 0=full, 1=stage+classify, 2=+hash, 3=+cache (no spill)
 template <int CACHE_N, bool GPOS, int TILE_N, int MODE = 0,
-          bool SPILL_ALL = false>  // SPILL_ALL: emit every word (no
-                                   // cache) — the inverted-index path
+          bool SPILL_ALL = false,  // emit every word (no cache)
+          bool COMPOSITE = false>  // keys = wordhash ^ splitmix64(doc)
+                                   // (doc from split_off binary search) —
+                                   // the inverted-index path, fusing the
+                                   // doc lookup + mix into the tokenizer
 __global__ __launch_bounds__(256) void tokenize_v6_kernel(
     const u8* __restrict__ text, long n, u64 pos_base,
     u64* __restrict__ tkeys, i64* __restrict__ tvals, u64* __restrict__ texm,
     u64 cap_mask, u64* __restrict__ out_hash, u64* __restrict__ out_pos,
     unsigned long long* __restrict__ spill_counter, long spill_cap,
-    unsigned long long* __restrict__ nwords, u64* __restrict__ cpos_g) {
+    unsigned long long* __restrict__ nwords, u64* __restrict__ cpos_g,
+    const i64* __restrict__ split_off = nullptr, int nsplit_off = 0,
+    long doc_base = 0) {
   __shared__ __align__(16) u8 tile[TILE_N + TOK_HALO];
   __shared__ u64 ckeys[CACHE_N];
   // exemplar positions: LDS normally; with GPOS a global side-buffer
@@ -776,6 +781,13 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
         len = (u32)(wlen > 0xFFFF ? 0xFFFF : wlen);
       }
       if (MODE == 2) { my_words += h; continue; }
+      if (COMPOSITE) {
+        // fuse the (word, doc) composite: doc from the split-offset
+        // table (L2-resident; ~10-step binary search per word)
+        i64 byte = (i64)(pos_base + (u64)(base + my0 + s));
+        int doc = ub_minus1(split_off, nsplit_off, byte);
+        h ^= splitmix64_dev((u64)(doc + doc_base));
+      }
       u64 k = remap_key(h);
       u64 p = ((pos_base + (u64)(base + my0 + s)) << 16) | (u64)len;
       if (SPILL_ALL) {

@@ -120,7 +120,35 @@ std::vector<torch::Tensor> tokenize_spill_v2(torch::Tensor text,
                        reinterpret_cast<unsigned long long*>(counter.data_ptr<i64>()),
                        cap,
                        reinterpret_cast<unsigned long long*>(nwords.data_ptr<i64>()),
-                       nullptr);
+                       nullptr, (const i64*)nullptr, 0, 0L);
+  }
+  return {out_hash, out_pos, counter, nwords};
+}
+
+std::vector<torch::Tensor> tokenize_spill_composite(
+    torch::Tensor text, long pos_base, long cap, torch::Tensor split_off,
+    long doc_base) {
+  TORCH_CHECK(text.is_cuda() && text.scalar_type() == torch::kUInt8 &&
+              text.is_contiguous(), "text must be contiguous u8 on GPU");
+  check_dev_i64(split_off, "split_off");
+  long n = text.numel();
+  auto opts = torch::TensorOptions().device(text.device()).dtype(torch::kInt64);
+  auto out_hash = torch::empty({cap}, opts);
+  auto out_pos = torch::empty({cap}, opts);
+  auto counter = torch::zeros({1}, opts);
+  auto nwords = torch::zeros({1}, opts);
+  if (n) {
+    auto dummy = torch::empty({16}, opts);
+    hipLaunchKernelGGL((tokenize_v6_kernel<16, false, 4096, 0, true, true>),
+                       dim3(grid_for(n, TOK_BYTES)), dim3(kBlock), 0,
+                       cur_stream(), text.data_ptr<u8>(), n, (u64)pos_base,
+                       u64p(dummy), dummy.data_ptr<i64>(), nullptr,
+                       (u64)15, u64p(out_hash), u64p(out_pos),
+                       reinterpret_cast<unsigned long long*>(counter.data_ptr<i64>()),
+                       cap,
+                       reinterpret_cast<unsigned long long*>(nwords.data_ptr<i64>()),
+                       nullptr, split_off.data_ptr<i64>(),
+                       (int)split_off.numel(), doc_base);
   }
   return {out_hash, out_pos, counter, nwords};
 }
@@ -205,7 +233,7 @@ void tokenize_cache_spill(
                          reinterpret_cast<unsigned long long*>(counter.data_ptr<i64>()),
                          spill_cap,
                          reinterpret_cast<unsigned long long*>(nwords.data_ptr<i64>()),
-                         cpg);
+                         cpg, (const i64*)nullptr, 0, 0L);
     } else if (!(v && v[0] == '1'))
       hipLaunchKernelGGL(tokenize_cache_spill_kernel,
                          dim3(grid_for(n, TOK_BYTES)), dim3(kBlock), 0,
@@ -261,7 +289,7 @@ double tok6_ablate(torch::Tensor text, long mode, long iters) {
                        reinterpret_cast<unsigned long long*>(ctr.data_ptr<i64>()),
                        cap,
                        reinterpret_cast<unsigned long long*>(nw.data_ptr<i64>()),
-                       u64p(cpg));
+                       u64p(cpg), (const i64*)nullptr, 0, 0L);
   };
   launch();  // warm
   hipEventRecord(e0, st);
@@ -569,6 +597,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("tokenize", &tokenize, "tokenize text -> (hash, pos, count)");
   m.def("tokenize_count", &tokenize_count,
         "fused tokenize + hash-table count");
+  m.def("tokenize_spill_composite", &tokenize_spill_composite,
+        "spill-all with fused (word,doc) composite keys");
   m.def("tokenize_spill_v2", &tokenize_spill_v2,
         "spill-all tokenizer (chunk-padded; filter HT_EMPTY)");
   m.def("tokenize_spill", &tokenize_spill,
