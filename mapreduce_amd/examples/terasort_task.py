@@ -1,0 +1,60 @@
+"""TeraSort as a six-function task script (host tier).
+
+Contract demo: the general engine sorts through the SAME machinery the
+reference uses — mapfn emits (key, payload), partitionfn range-partitions
+keys, the engine's sorted spills + k-way merge (job.lua:194,
+utils.lua:206-271 semantics) deliver globally sorted output (partition
+order x in-partition key order).  reducefn is identity; no combiner.
+
+The GPU-scale path is mapreduce_amd.gpu.terasort (radix sort + xGMI
+exchange); this script is the API-parity form for arbitrary sizes that
+need no GPU.
+
+init_args: {"n": total keys, "splits": map jobs, "parts": partitions,
+"seed": int}.
+"""
+
+from __future__ import annotations
+
+import random
+
+_CFG = {"n": 10_000, "splits": 8, "parts": 4, "seed": 0}
+RESULTS = []  # finalfn: [(key, payload)] globally sorted
+
+KEY_SPACE = 1 << 32
+
+
+def init(arg):
+    if arg:
+        _CFG.update({k: v for k, v in arg.items() if k in _CFG})
+
+
+def taskfn(emit):
+    per = _CFG["n"] // _CFG["splits"]
+    for s in range(_CFG["splits"]):
+        emit(s + 1, {"seed": _CFG["seed"] * 1000 + s, "count": per})
+
+
+def mapfn(key, value, emit):
+    rng = random.Random(value["seed"])
+    for i in range(value["count"]):
+        k = rng.randrange(KEY_SPACE)
+        emit(k, (key, i))  # payload records provenance
+
+
+def partitionfn(key):
+    # contiguous key ranges -> partition-major order IS global order
+    return int(key * _CFG["parts"] // KEY_SPACE)
+
+
+def reducefn(key, values, emit):
+    for v in values:
+        emit(v)
+
+
+def finalfn(pairs):
+    RESULTS.clear()
+    for key, values in pairs:
+        for v in values:
+            RESULTS.append((key, tuple(v)))
+    return True

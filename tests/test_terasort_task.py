@@ -1,0 +1,34 @@
+"""Host-tier TeraSort task script: sorted-output guarantee through the
+general engine (the reference's partition-file-then-key ordering,
+server.lua:360-385)."""
+
+import importlib
+import random
+
+
+def test_terasort_task_globally_sorted():
+    import mapreduce_amd.examples.terasort_task as ts
+    from mapreduce_amd import job as jobmod, run_local
+    importlib.reload(ts)
+    jobmod._module_cache.clear()
+    jobmod._inited.clear()
+
+    fns = {r: ts for r in ("taskfn", "mapfn", "partitionfn", "reducefn",
+                           "finalfn")}
+    srv = run_local({"fns": fns, "verbose": False,
+                     "init_args": {"n": 4000, "splits": 5, "parts": 4,
+                                   "seed": 3}},
+                    nworkers=3)
+    assert srv.finished
+    keys = [k for k, _ in ts.RESULTS]
+    assert len(keys) == 4000
+    assert keys == sorted(keys)  # global order
+    # content parity vs re-generating the same records
+    exp = []
+    per = 4000 // 5
+    for s in range(5):
+        rng = random.Random(3 * 1000 + s)
+        for i in range(per):
+            exp.append((rng.randrange(1 << 32), (str(s + 1), i)))
+    exp.sort()
+    assert sorted(ts.RESULTS) == exp
