@@ -1,0 +1,96 @@
+"""StoreCoordinator over a real TCPStore (in-process master + client):
+CAS claims, counters, error channel, namespace drop."""
+
+import socket
+import threading
+
+import pytest
+
+from mapreduce_amd.parallel.coord import StoreCoordinator
+from mapreduce_amd.task import Task, make_job
+from mapreduce_amd.utils import STATUS, TASK_STATUS
+
+
+def free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+@pytest.fixture()
+def pair():
+    port = free_port()
+    master = StoreCoordinator(f"tcp://127.0.0.1:{port}", db="t",
+                              listen=True, timeout_s=20)
+    client = StoreCoordinator(f"tcp://127.0.0.1:{port}", db="t",
+                              listen=False, timeout_s=20)
+    return master, client
+
+
+def test_doc_roundtrip_and_cas(pair):
+    m, c = pair
+    m.set_doc("x", {"a": 1})
+    doc, raw = c.get_doc("x")
+    assert doc == {"a": 1}
+    assert c.cas_doc("x", raw, {"a": 2})
+    assert not m.cas_doc("x", raw, {"a": 3})  # stale token loses
+    doc2, _ = m.get_doc("x")
+    assert doc2 == {"a": 2}
+    # create-if-absent
+    assert m.cas_doc("fresh", None, {"v": 1})
+    assert not c.cas_doc("fresh", None, {"v": 2})
+
+
+def test_counters_and_errors(pair):
+    m, c = pair
+    assert m.add("n", 5) == 5
+    assert c.add("n", 2) == 7
+    c.insert_error("w1", "boom")
+    m.insert_error("w2", "bang")
+    errs, hi = m.get_errors(0)
+    assert [e["who"] for e in errs] == ["w1", "w2"]
+    errs2, _ = m.get_errors(hi)
+    assert errs2 == []
+
+
+def test_claims_race_over_tcp(pair):
+    m, c = pair
+    t = Task(m)
+    t.create_collection(TASK_STATUS.MAP, {
+        "fns": {}, "storage": "mem:tcp", "result_ns": "result"}, 1)
+    t.insert_jobs(Task.MAP_JOBS, [make_job(str(i), i) for i in range(20)])
+
+    wins = []
+    lock = threading.Lock()
+
+    def claimer(coord, name):
+        tk = Task(coord)
+        tk.update()
+        while True:
+            ns, doc = tk.take_next_job(name, name)
+            if doc is None:
+                return
+            with lock:
+                wins.append(doc["_id"])
+
+    th1 = threading.Thread(target=claimer, args=(m, "a"))
+    th2 = threading.Thread(target=claimer, args=(c, "b"))
+    th1.start()
+    th2.start()
+    th1.join()
+    th2.join()
+    assert sorted(wins) == sorted(str(i) for i in range(20))
+    assert len(set(wins)) == 20  # exactly-once over TCP CAS
+
+
+def test_drop_ns(pair):
+    m, c = pair
+    m.set_doc("map_jobs/1", {"x": 1})
+    m.set_ids("map_jobs", ["1"])
+    assert c.get_ids("map_jobs") == ["1"]
+    c.drop_ns("map_jobs")
+    assert m.get_ids("map_jobs") == []
+    doc, _ = m.get_doc("map_jobs/1")
+    assert doc is None
