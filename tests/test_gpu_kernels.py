@@ -259,6 +259,19 @@ def test_streaming_wordcount_mode_vs_counter(dev):
     assert dict(res2.to_host()) == got
 
 
+def test_wordcount_deterministic_across_runs(dev):
+    """Counts are exactly reproducible run-to-run despite nondeterministic
+    kernel scheduling (aggregation is commutative; exemplars may differ in
+    POSITION but always name the same word)."""
+    from mapreduce_amd.gpu.corpus import make_corpus
+    from mapreduce_amd.gpu.wordcount import WordCountJob
+    c = make_corpus(dev, nwords=200_000, nsplits=7, vocab_size=3000, seed=6)
+    job = WordCountJob(dev, vocab_estimate=6000)
+    r1 = dict(job.run(c.text, c.splits()).to_host())
+    r2 = dict(job.run(c.text, c.splits()).to_host())
+    assert r1 == r2
+
+
 def test_cluster_runner_dynamic_gpu(dev):
     """Control-plane dynamic claims driving GPU map jobs (per-job CAS,
     WRITTEN transitions) — single rank on hardware."""
