@@ -61,53 +61,63 @@ class StreamLoader:
         self._copy_stream = (torch.cuda.Stream(self.device)
                              if self.device.type == "cuda" else None)
 
-    def _read_chunks(self) -> Iterator[Tuple[bytes, int]]:
+    def __iter__(self) -> Iterator[Tuple[torch.Tensor, int]]:
+        """Zero-copy-ish pipeline: file.readinto(pinned ring buffer) ->
+        async H2D on the side stream; the consumer's stream waits on the
+        copy event, and a buffer is reused only after its previous copy
+        completed."""
+        dev = self.device
+        use_cuda = dev.type == "cuda"
         size = os.path.getsize(self.path)
+        bufs = [torch.empty(self.chunk_bytes, dtype=torch.uint8,
+                            pin_memory=use_cuda) for _ in range(2)]
+        buf_ev: list = [None, None]
+        pending: Optional[Tuple[torch.Tensor, int,
+                                Optional["torch.cuda.Event"]]] = None
         with open(self.path, "rb") as fh:
             base = 0
             carry = b""
-            while base + len(carry) < size or carry:
-                want = self.chunk_bytes - len(carry)
-                data = carry + fh.read(want)
-                if not data:
-                    return
-                if base + len(data) < size:
+            bi = 0
+            while base < size or carry:
+                if buf_ev[bi] is not None:
+                    buf_ev[bi].synchronize()  # previous copy out of buffer
+                    buf_ev[bi] = None
+                buf = bufs[bi]
+                mv = memoryview(buf.numpy())
+                nc = len(carry)
+                mv[:nc] = carry
+                nread = fh.readinto(mv[nc:])
+                total = nc + nread
+                if total == 0:
+                    break
+                at_eof = (base + total) >= size
+                if not at_eof:
                     # cut at the last whitespace so no word spans chunks
-                    cut = max(data.rfind(b" "), data.rfind(b"\n"),
-                              data.rfind(b"\t"))
-                    if cut <= 0:
-                        cut = len(data)  # one giant word: hand it over whole
-                    else:
-                        cut += 1
+                    tail0 = max(0, total - 65536)
+                    tail = bytes(mv[tail0:total])
+                    cut = max(tail.rfind(b" "), tail.rfind(b"\n"),
+                              tail.rfind(b"\t"))
+                    cut = total if cut < 0 else tail0 + cut + 1
                 else:
-                    cut = len(data)
-                yield data[:cut], base
-                carry = data[cut:]
+                    cut = total
+                carry = bytes(mv[cut:total])
+                if use_cuda:
+                    with torch.cuda.stream(self._copy_stream):
+                        d = buf[:cut].to(dev, non_blocking=True)
+                        ev = torch.cuda.Event()
+                        ev.record(self._copy_stream)
+                    buf_ev[bi] = ev
+                else:
+                    d = buf[:cut].clone()
+                    ev = None
+                if pending is not None:
+                    pd, pb, pev = pending
+                    if pev is not None:
+                        torch.cuda.current_stream(dev).wait_event(pev)
+                    yield pd, pb
+                pending = (d, base, ev)
                 base += cut
-                if base >= size and not carry:
-                    return
-
-    def __iter__(self) -> Iterator[Tuple[torch.Tensor, int]]:
-        dev = self.device
-        use_cuda = dev.type == "cuda"
-        pending: Optional[Tuple[torch.Tensor, int, Optional[torch.cuda.Event]]] = None
-        for raw, base in self._read_chunks():
-            host = torch.frombuffer(bytearray(raw), dtype=torch.uint8)
-            if use_cuda:
-                host = host.pin_memory()
-                with torch.cuda.stream(self._copy_stream):
-                    d = host.to(dev, non_blocking=True)
-                    ev = torch.cuda.Event()
-                    ev.record(self._copy_stream)
-            else:
-                d = host
-                ev = None
-            if pending is not None:
-                pd, pb, pev = pending
-                if pev is not None:
-                    torch.cuda.current_stream(dev).wait_event(pev)
-                yield pd, pb
-            pending = (d, base, ev)
+                bi ^= 1
         if pending is not None:
             pd, pb, pev = pending
             if pev is not None:
