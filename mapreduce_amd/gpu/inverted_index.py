@@ -6,12 +6,16 @@ is a group-by — with the sort-once design the doc list of a key IS its
 contiguous segment, so reduce is segment bookkeeping, not computation
 (SURVEY.md K4 -> K1+K5).
 
-Pipeline per rank:
-  tokenize_spill -> (hash, pos); doc id = searchsorted(split offsets, pos)
-  composite sort: LSD radix by doc (low bits) then stable by hash
-  segment by (hash, doc) -> per-(word,doc) term frequency
-  [world > 1] exchange by hash partition (mulhi), re-sort, re-segment
-  segment by hash -> doc-list offsets per word + exemplar bytes
+Pipeline per rank (GPU):
+  tokenize_spill_composite -> (wordhash ^ splitmix64(doc), pos) with the
+    doc id binary-searched from the split offsets IN-KERNEL
+  bucketize + per-bucket LDS count -> unique (word, doc) postings with
+    term frequencies (aggregation BEFORE any sort: ~7x fewer elements)
+  recover (doc, wordhash) from exemplar positions; sort by doc bits then
+    stable by wordhash -> doc lists ordered within each word
+  [world > 1] exchange by wordhash partition (mulhi), re-sort, re-merge
+  segment by wordhash -> doc-list offsets per word + exemplar bytes
+(CPU tier keeps the sort-first reference shape.)
 """
 
 from __future__ import annotations
