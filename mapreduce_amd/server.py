@@ -72,6 +72,7 @@ class Server:
         }
         self.poll_interval = params.get("poll_interval", DEFAULT_SLEEP)
         self.heartbeat_timeout = params.get("heartbeat_timeout")
+        self.stall_timeout = params.get("stall_timeout")
         self.verbose = params.get("verbose", True)
         self.fns = FnSet(fns, self.params["init_args"])
         self.fs = fsmod.router(storage, self.params["path"])
@@ -158,6 +159,7 @@ class Server:
         promote exhausted BROKEN jobs to FAILED, requeue stale RUNNING jobs
         (heartbeat), print % progress, drain the error channel."""
         last_pct = -1
+        last_progress = (-1, gettime())
         while True:
             self.task.promote_broken(ns)
             if self.heartbeat_timeout:
@@ -169,6 +171,19 @@ class Server:
             for e in errors:
                 self._log(f"worker error [{e['who']}]: {e['msg']}")
             written, failed, total = self.task.count_done(ns)
+            if written + failed != last_progress[0]:
+                last_progress = (written + failed, gettime())
+            elif (self.stall_timeout
+                  and gettime() - last_progress[1] > self.stall_timeout):
+                # no progress for stall_timeout: the worker pool is likely
+                # depleted (each worker quits after MAX_WORKER_RETRIES
+                # distinct failures) — force-fail the stuck BROKEN jobs so
+                # the task completes with a failure count instead of
+                # polling forever (liveness fix over the reference)
+                n = self.task.force_fail_incomplete(ns)
+                if n:
+                    self._log(f"stall: force-failed {n} {phase} jobs")
+                last_progress = (written + failed, gettime())
             if total:
                 pct = int(100 * (written + failed) / total)
                 if pct != last_pct:

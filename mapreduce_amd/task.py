@@ -192,6 +192,26 @@ class Task:
                     break
         return n
 
+    def force_fail_incomplete(self, ns: str) -> int:
+        """Promote every WAITING or BROKEN job to FAILED — the server's
+        stall-timeout escape hatch when the worker pool is depleted
+        (liveness addition over the reference).  RUNNING jobs are left
+        alone: a live worker holds them (dead holders are the heartbeat
+        timeout's case)."""
+        n = 0
+        for i in self.coord.get_ids(ns):
+            while True:
+                doc, raw = self.coord.get_doc(f"{ns}/{i}")
+                if doc is None or doc["status"] not in (STATUS.WAITING,
+                                                        STATUS.BROKEN):
+                    break
+                new = dict(doc)
+                new["status"] = STATUS.FAILED
+                if self.coord.cas_doc(f"{ns}/{i}", raw, new):
+                    n += 1
+                    break
+        return n
+
     def requeue_stale(self, ns: str, timeout_s: float) -> int:
         """Liveness repair the reference lacks (SURVEY.md §5 'a dead worker's
         RUNNING job is not auto-requeued'): RUNNING jobs older than

@@ -123,6 +123,20 @@ def test_worker_gives_up_after_repeated_failures():
     assert len(broken) >= MAX_JOB_RETRIES
 
 
+def test_stall_timeout_force_fails_with_depleted_pool():
+    """Every worker exhausts its retry budget on always-crashing jobs; the
+    reference would poll forever — the stall timeout completes the task
+    with FAILED jobs instead (liveness addition)."""
+    def mapfn(key, value, emit):
+        raise RuntimeError("always broken")
+
+    fns = dict(WC_FNS, mapfn=mapfn)
+    srv = run_local({"fns": allroles(fns), "verbose": False,
+                     "stall_timeout": 0.5}, nworkers=2)
+    assert srv.finished
+    assert srv.stats["map_failed"] == 4
+
+
 def test_error_channel_reaches_server(capsys):
     """Worker tracebacks flow through the error channel to the server log
     (cnn.lua:62-78 -> server.lua:219-228)."""
