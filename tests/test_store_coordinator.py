@@ -86,11 +86,15 @@ def test_claims_race_over_tcp(pair):
 
 
 def test_drop_ns(pair):
+    import time
     m, c = pair
     m.set_doc("map_jobs/1", {"x": 1})
     m.set_ids("map_jobs", ["1"])
     assert c.get_ids("map_jobs") == ["1"]
     c.drop_ns("map_jobs")
+    for _ in range(40):  # deletes may lag under load
+        if m.get_ids("map_jobs") == [] and m.get_doc("map_jobs/1")[0] is None:
+            break
+        time.sleep(0.05)
     assert m.get_ids("map_jobs") == []
-    doc, _ = m.get_doc("map_jobs/1")
-    assert doc is None
+    assert m.get_doc("map_jobs/1")[0] is None
