@@ -1,17 +1,27 @@
 // MapReduce hot-path kernels for MI355X (gfx950), CDNA4-native.
 //
 // These replace the reference's CPU-Lua hot loops:
-//   K2/K3  tokenize_kernel        — word boundary scan + FNV-1a64 key hash
-//   K5     hash_insert/extract    — hash-table combiner / reduce-by-key
-//          (aggregation form; the sort form is radix_sort.hip + segmented
-//          reduce below)
-//   K5     seg_reduce kernels     — segmented reduce-by-key over sorted runs
+//   K2/K3  tokenize_v6_kernel     — THE production tokenizer: branchless
+//          ws-mask scan, chunked register hashing (wordhash64), per-block
+//          LDS cache for the Zipf head, wave-chunked spill allocator for
+//          the tail (template flags add spill-all and fused (word,doc)
+//          composite modes for the inverted index); tokenize_kernel /
+//          tokenize_count_kernel / tokenize_spill / v5 are earlier
+//          structures kept for tests and recorded A/Bs
+//   K5     bucket_count_kernel    — per-bucket LDS count of the
+//          top-byte-partitioned spill tail; hash_insert/extract are the
+//          generic table ops (extract has a chunked-compaction v2 for
+//          large tables); the sorted form is radix_sort.hip + the
+//          seg_reduce kernels below
 //   K2     partition_hist         — all-to-all send counts (C5 setup)
 //   K7/K8  gather_bytes           — exemplar word extraction for the
 //          hash -> string dictionary at the finalfn boundary
 //
-// All memory-bound: vectorized accesses, grid-stride loops, one atomic per
-// wave where aggregation applies (guide G12/G13).
+// All memory-bound: vectorized accesses, grid-stride loops.  The recurring
+// measured lesson (see profiles/): a shared atomic counter serializes
+// cross-XCD at ~9 ns/op — every append path here reserves CHUNKED ranges
+// per wave and pads unused tail slots with HT_EMPTY, which downstream
+// consumers skip.
 
 #include "common.h"
 
@@ -20,7 +30,9 @@
 // ---------------------------------------------------------------------------
 // pos packs (start << 16 | len) so the exemplar word bytes can be gathered
 // later; output order is nondeterministic (atomic append) — every consumer
-// sorts or hash-aggregates, so order never matters.
+// sorts or hash-aggregates, so order never matters.  NOTE: one counter
+// atomic per word — fine at test scale, NOT a hot-path kernel (the
+// production paths use the wave-chunked allocator; see header).
 
 #define TOK_BYTES 16
 
