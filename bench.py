@@ -109,6 +109,8 @@ def main() -> int:
     if rank == 0:
         out = {
             "metric": "words/sec",
+            "metric_detail": "words/sec (whole node) Europarl word-count "
+                             "at N MI355X; job wall-clock (BASELINE.json)",
             "value": value,
             "unit": "words/s",
             "n_gpus": world,

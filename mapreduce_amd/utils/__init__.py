@@ -145,6 +145,10 @@ def assert_check(value: Any, path: str = "value") -> None:
 # (utils.lua:100-120 writer, utils.lua:222-224 load() parser).  A record is a
 # length-prefixed pickle of (key, values_list); files are streams of records
 # sorted by sort_key(key).
+#
+# Trust model: spill files are produced and consumed inside one trusted
+# cluster, exactly like the reference's load(line)() evaluation of spill
+# rows (utils.lua:222-224) — neither format is safe for untrusted input.
 _LEN = struct.Struct("<I")
 
 
