@@ -210,6 +210,8 @@ void tokenize_cache_spill(
       int tsz = ts ? atoi(ts) : 4096;
       auto kfn = tokenize_v6_kernel<2048, false, 4096>;
       if (gpos && tsz == 8192) kfn = tokenize_v6_kernel<2048, true, 8192>;
+      else if (gpos && cache == 4096) kfn = tokenize_v6_kernel<4096, true, 4096>;
+      else if (gpos && cache == 1024) kfn = tokenize_v6_kernel<1024, true, 4096>;
       else if (gpos) kfn = tokenize_v6_kernel<2048, true, 4096>;
       else if (cache == 512) kfn = tokenize_v6_kernel<512, false, 4096>;
       else if (cache == 1024) kfn = tokenize_v6_kernel<1024, false, 4096>;
@@ -217,7 +219,7 @@ void tokenize_cache_spill(
       static torch::Tensor cpos_g;  // persistent side-buffer (GPOS only)
       u64* cpg = nullptr;
       if (gpos) {
-        long need = blocks * 2048;
+        long need = blocks * (cache > 2048 ? cache : 2048);
         if (!cpos_g.defined() || cpos_g.numel() < need ||
             cpos_g.device() != text.device())
           cpos_g = torch::empty({need},
