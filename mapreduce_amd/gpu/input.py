@@ -92,12 +92,21 @@ class StreamLoader:
                     break
                 at_eof = (base + total) >= size
                 if not at_eof:
-                    # cut at the last whitespace so no word spans chunks
+                    # cut at the last whitespace so no word spans chunks:
+                    # scan a 64 KB tail first, the whole chunk if the tail
+                    # is one ws-free run (only a word longer than the
+                    # chunk itself is unsplittable)
                     tail0 = max(0, total - 65536)
                     tail = bytes(mv[tail0:total])
                     cut = max(tail.rfind(b" "), tail.rfind(b"\n"),
                               tail.rfind(b"\t"))
-                    cut = total if cut < 0 else tail0 + cut + 1
+                    if cut >= 0:
+                        cut = tail0 + cut + 1
+                    else:
+                        whole = bytes(mv[:total])
+                        cut = max(whole.rfind(b" "), whole.rfind(b"\n"),
+                                  whole.rfind(b"\t"))
+                        cut = total if cut < 0 else cut + 1
                 else:
                     cut = total
                 carry = bytes(mv[cut:total])

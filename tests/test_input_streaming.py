@@ -53,6 +53,23 @@ def test_stream_loader_chunks_exact(tmp_path):
     assert got == exp
 
 
+def test_stream_loader_long_wsfree_tail(tmp_path):
+    """A chunk whose last 64 KB holds no whitespace must still cut at a
+    word boundary (full-chunk fallback scan)."""
+    p = tmp_path / "nasty.txt"
+    # 100 KB ws-free run (> the 64 KB tail scan, < the 256 KB chunk)
+    # followed by enough text that the run lands inside a chunk tail
+    data = (b"alpha beta " * 12000) + (b"x" * 100_000) + \
+        (b" tail end " * 4000)
+    p.write_bytes(data)
+    # chunk ends at 200,000 — inside the x-run with its whole 64 KB tail
+    # ws-free, so the cut must come from the full-chunk fallback scan
+    got = collections.Counter()
+    for chunk, base in StreamLoader(str(p), "cpu", chunk_bytes=200_000):
+        got.update(bytes(chunk.numpy().tobytes()).split())
+    assert got == collections.Counter(data.split())
+
+
 def test_streamed_wordcount_job(tmp_path):
     """Chunks feed map jobs one by one (begin_map once, map over chunks
     via per-chunk text) — counts equal the whole-file oracle."""
