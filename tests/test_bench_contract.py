@@ -33,6 +33,23 @@ def run_bench(extra, env=None):
     return json.loads(line)
 
 
+def run_torchrun_bench(nproc, bench_args):
+    """torchrun rendezvous can transiently fail under suite load (the
+    probed free port may be reclaimed) — retry once with a fresh port."""
+    last = None
+    for _ in range(2):
+        port = free_port()
+        try:
+            return run_bench([
+                "-m", "torch.distributed.run", "--nnodes=1",
+                "--nproc-per-node", str(nproc),
+                "--master-addr", "127.0.0.1", "--master-port", str(port),
+                "bench.py"] + bench_args)
+        except AssertionError as e:  # pragma: no cover - flake path
+            last = e
+    raise last
+
+
 def test_bench_single_process_contract():
     d = run_bench(["bench.py", "--steps", "2", "--warmup", "1",
                    "--words", "20000", "--splits", "4", "--vocab", "500",
@@ -48,27 +65,21 @@ def test_bench_single_process_contract():
     assert d["config"]["parallelism"] == "dp1"
 
 
-@pytest.mark.timeout(300)
+@pytest.mark.timeout(500)
 def test_bench_torchrun_ws4_contract():
     """4-rank canary for the driver's N=4/8 scaling runs."""
-    port = free_port()
-    d = run_bench([
-        "-m", "torch.distributed.run", "--nnodes=1", "--nproc-per-node", "4",
-        "--master-addr", "127.0.0.1", "--master-port", str(port),
-        "bench.py", "--gpus", "4", "--steps", "2", "--warmup", "1",
+    d = run_torchrun_bench(4, [
+        "--gpus", "4", "--steps", "2", "--warmup", "1",
         "--words", "12000", "--splits", "3", "--vocab", "400",
         "--device", "cpu"])
     assert d["n_gpus"] == 4
     assert d["config"]["global_batch"] == 48000
 
 
-@pytest.mark.timeout(240)
+@pytest.mark.timeout(500)
 def test_bench_torchrun_ws2_contract():
-    port = free_port()
-    d = run_bench([
-        "-m", "torch.distributed.run", "--nnodes=1", "--nproc-per-node", "2",
-        "--master-addr", "127.0.0.1", "--master-port", str(port),
-        "bench.py", "--gpus", "2", "--steps", "2", "--warmup", "1",
+    d = run_torchrun_bench(2, [
+        "--gpus", "2", "--steps", "2", "--warmup", "1",
         "--words", "20000", "--splits", "4", "--vocab", "500",
         "--device", "cpu"])
     assert d["n_gpus"] == 2
