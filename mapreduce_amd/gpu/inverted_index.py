@@ -159,7 +159,7 @@ class InvertedIndexJob:
             torch.cumsum(totals, 0, out=bucket_off[1:])
             table = ops.make_table(max(1 << 16, k2.numel() // 6), dev)
             ops.ext().bucket_count(hk, pv, bucket_off, 256, 32,
-                                   table.tkeys, table.tvals, table.texm)
+                                   table.tkeys, table.tvals, table.texm, 0)
             uk2, tf, upos = table.extract()
             ud = torch.searchsorted(starts, upos >> 16, right=True) - 1
             ud = ud + self.doc_base

@@ -143,10 +143,40 @@ class WordCountJob:
         self.reset()
         self._text = text
         if self.mode == "streaming":
-            # + slack for the wave-chunked spill allocator's padded chunk
-            # tails (<= 2048 blocks x 4 waves x 512-entry chunks)
-            cap = text.numel() // 2 + 16 + 2048 * 4 * 512
+            import os
             opts = dict(dtype=torch.int64, device=self.device)
+            # bucketed direct spill (MR_TOK_BSPILL=1): the tokenizer lands
+            # misses straight into 256 per-top-byte-bucket regions,
+            # removing the radix_pass(56) bucketize and all pad entries.
+            # MEASURED 5x SLOWER (11.1 vs 2.09 ms/step): one reservation
+            # atomic per distinct bucket per wave-window exposes a full
+            # L2-atomic round trip per miss-group (~430 serial stalls per
+            # wave), exactly the latency the 512-entry chunked allocator
+            # amortizes away.  Kept env-gated for the record; default OFF.
+            self._bspill = (
+                self.device.type == "cuda"
+                and os.environ.get("MR_TOK_BSPILL", "0") == "1"
+                and os.environ.get("MR_TOK_CACHE", "2048") == "2048"
+                and os.environ.get("MR_TOK_TILE", "4096") == "4096"
+                and os.environ.get("MR_TOKENIZE_V4", "0") != "1"
+                and os.environ.get("MR_TOKENIZE_V5", "0") != "1")
+            if self._bspill:
+                # same total footprint as the chunked layout; uniform hash
+                # top byte spreads misses, ~30x headroom per bucket, loud
+                # overflow check in finish_map
+                bcap = max(4096, (text.numel() // 2 + 16) // 256 + 2048)
+                self._spill_h = torch.empty(256 * bcap, **opts)
+                self._spill_p = torch.empty(256 * bcap, **opts)
+                self._spill_c = torch.zeros(256, **opts)
+                self._spill_bcap = bcap
+                return
+            # + slack for the wave-chunked spill allocator's padded chunk
+            # tails (<= 2048 blocks x 4 waves x one chunk each)
+            try:
+                schunk = int(os.environ.get("MR_SPILL_CHUNK", "2048"))
+            except ValueError:
+                schunk = 2048
+            cap = text.numel() // 2 + 16 + 2048 * 4 * max(schunk, 512)
             self._spill_h = torch.empty(cap, **opts)
             self._spill_p = torch.empty(cap, **opts)
             self._spill_c = torch.zeros(1, **opts)
@@ -157,10 +187,16 @@ class WordCountJob:
         Streaming mode appends cache misses to the shared spill arrays
         (atomic counter append composes across launches)."""
         if self.mode == "streaming":
-            ops.ext().tokenize_cache_spill(
-                self._text[s:e], s, self.table.tkeys, self.table.tvals,
-                self.table.texm, self._spill_cap, self._nwords,
-                self._spill_h, self._spill_p, self._spill_c)
+            if self._bspill:
+                ops.ext().tokenize_cache_spill_bucketed(
+                    self._text[s:e], s, self.table.tkeys, self.table.tvals,
+                    self.table.texm, self._spill_bcap, self._nwords,
+                    self._spill_h, self._spill_p, self._spill_c)
+            else:
+                ops.ext().tokenize_cache_spill(
+                    self._text[s:e], s, self.table.tkeys, self.table.tvals,
+                    self.table.texm, self._spill_cap, self._nwords,
+                    self._spill_h, self._spill_p, self._spill_c)
         else:
             self.table.tokenize_count(self._text[s:e], s, self._nwords)
 
@@ -168,6 +204,25 @@ class WordCountJob:
         """Drain the spill through bucketize + per-bucket LDS count.
         Returns the word count (the one host sync of the map phase)."""
         if self.mode == "streaming":
+            import os
+            slices = int(os.environ.get("MR_BUCKET_SLICES", "32"))
+            if self._bspill:
+                # misses are already bucket-partitioned in per-bucket
+                # regions; counters hold exact per-bucket lengths
+                cnts = self._spill_c.cpu()
+                n = int(self._nwords.item())
+                mx = int(cnts.max().item())
+                if mx > self._spill_bcap:
+                    raise RuntimeError(
+                        f"bucketed spill overflow: {mx} reserved > "
+                        f"per-bucket cap {self._spill_bcap}")
+                if int(cnts.sum().item()):
+                    ops.ext().bucket_count(
+                        self._spill_h, self._spill_p, self._spill_c, 256,
+                        slices, self.table.tkeys, self.table.tvals,
+                        self.table.texm, self._spill_bcap)
+                self._spill_h = self._spill_p = None
+                return n
             nspill = int(self._spill_c.item())
             n = int(self._nwords.item())
             if nspill > self._spill_cap:
@@ -184,15 +239,13 @@ class WordCountJob:
                 bucket_off = torch.zeros(257, dtype=torch.int64,
                                          device=self.device)
                 torch.cumsum(totals, 0, out=bucket_off[1:])
-                import os
                 # 32 slices/bucket: the HT_EMPTY chunk padding all lands
                 # in bucket 255 (top byte 0xFF), so fine slicing keeps its
                 # blocks off the critical path (sweep: 8=3.19, 16=2.58,
                 # 32=2.38, 64=2.40 ms/step)
-                slices = int(os.environ.get("MR_BUCKET_SLICES", "32"))
                 ops.ext().bucket_count(hk, pv, bucket_off, 256, slices,
                                        self.table.tkeys, self.table.tvals,
-                                       self.table.texm)
+                                       self.table.texm, 0)
             self._spill_h = self._spill_p = None
             return n
         return int(self._nwords.item())

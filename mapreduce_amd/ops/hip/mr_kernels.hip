@@ -655,10 +655,20 @@ __global__ __launch_bounds__(256) void tokenize_v5_kernel(
 // MODE (ablation, §5.4 rule 8): 0=full, 1=stage+classify, 2=+hash, 3=+cache (no spill)
 template <int CACHE_N, bool GPOS, int TILE_N, int MODE = 0,
           bool SPILL_ALL = false,  // emit every word (no cache)
-          bool COMPOSITE = false>  // keys = wordhash ^ splitmix64(doc)
+          bool COMPOSITE = false,  // keys = wordhash ^ splitmix64(doc)
                                    // (doc from split_off binary search) —
                                    // the inverted-index path, fusing the
                                    // doc lookup + mix into the tokenizer
+          int SCHUNK = 512,  // chunked-allocator reservation size: pads
+                             // scale with it (waves x SCHUNK/2 wasted
+                             // tail entries) vs atomic amortization
+          int NBKT = 0>  // >0: bucketed direct spill — misses go straight
+                         // into per-top-byte-bucket regions (out[b*cap ..])
+                         // via wave-cooperative reservation on NBKT
+                         // counters, replacing the single-array spill +
+                         // the later radix_pass(56) bucketize entirely
+                         // (that pass measured ~320 us/step: hist 63 +
+                         // scatter 255, profiles/kernel_stats_final_step)
 __global__ __launch_bounds__(256) void tokenize_v6_kernel(
     const u8* __restrict__ text, long n, u64 pos_base,
     u64* __restrict__ tkeys, i64* __restrict__ tvals, u64* __restrict__ texm,
@@ -687,7 +697,7 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
   // 3.2 ms kernel (ablation: no-spill 0.48 ms vs full 3.19 ms with
   // near-identical static code).  Chunk tails are padded with HT_EMPTY
   // keys; downstream consumers skip them.
-  constexpr int SPILL_CHUNK = 512;  // >= max spills per wave-window (64x8)
+  constexpr int SPILL_CHUNK = SCHUNK;
   long wchunk = -1;
   int wleft = 0;
   long tile0 = (long)blockIdx.x * TILE_N;
@@ -842,9 +852,48 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
       }
     }
     // window-level spill reservation: every lane of the wave is present
-    // here (wactive design above) — wave-prefix the miss counts, one
-    // global atomic per wave per window, unrolled masked stores
-    {
+    // here (wactive design above)
+    if (NBKT > 0) {
+      // bucketed direct spill: for each word slot round, group the wave's
+      // misses by top-byte bucket and reserve per-bucket slots with ONE
+      // atomic per distinct bucket per round.  The NBKT distinct counter
+      // addresses spread the atomic traffic (the same-address serialization
+      // that motivated the chunked allocator does not apply), writes of a
+      // round's same-bucket members are adjacent, and no pad entries exist
+      // — counters are exact per-bucket lengths.
+      #pragma unroll
+      for (int wi = 0; wi < 8; ++wi) {
+        unsigned long long pending =
+            __ballot((miss_mask >> wi) & 1u);
+        if (__ballot(miss_mask >> wi) == 0) break;  // no lane has more slots
+        bool mine = (miss_mask >> wi) & 1u;
+        u64 bk = mine ? (sh_[wi] >> 56) % (u64)NBKT : 0;
+        int lane = threadIdx.x & (WAVE - 1);
+        unsigned long long lt_mask =
+            (lane == 63) ? ~0ull >> 1 : ((1ull << lane) - 1);
+        while (pending) {
+          int leader = __ffsll(pending) - 1;
+          u64 lead_bk = __shfl(bk, leader, WAVE);
+          unsigned long long members =
+              __ballot(mine && bk == lead_bk) & pending;
+          int nmem = __popcll(members);
+          unsigned long long base = 0;
+          if (lane == leader)
+            base = atomicAdd(&spill_counter[lead_bk],
+                             (unsigned long long)nmem);
+          base = __shfl(base, leader, WAVE);
+          if (members & (1ull << lane)) {
+            long r = (long)base + __popcll(members & lt_mask);
+            if (r < spill_cap) {  // overflow: counter runs past cap — the
+              long o = (long)lead_bk * spill_cap + r;  // host detects
+              out_hash[o] = sh_[wi];                   // max(cnt) > cap
+              out_pos[o] = sp_[wi];
+            }
+          }
+          pending &= ~members;
+        }
+      }
+    } else {
       u32 my_ns = (u32)__popc(miss_mask);
       u32 incl = my_ns;
       #pragma unroll
@@ -855,14 +904,18 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
       u32 wave_total = __shfl(incl, WAVE - 1, WAVE);
       int lane = threadIdx.x & (WAVE - 1);
       if (wave_total && (int)wave_total > wleft) {
-        // retire the old chunk's tail (pad with HT_EMPTY) + grab a new one
+        // retire the old chunk's tail (pad with HT_EMPTY) + grab a new
+        // one (at least a window's worth: a pathological window can
+        // produce up to 64x8 misses, more than a small SPILL_CHUNK)
         for (int i = lane; i < wleft; i += WAVE)
           if (wchunk + i < spill_cap) out_hash[wchunk + i] = HT_EMPTY;
+        int grab = (int)wave_total > SPILL_CHUNK ? (int)wave_total
+                                                 : SPILL_CHUNK;
         unsigned long long nb = 0;
         if (lane == 0)
-          nb = atomicAdd(spill_counter, (unsigned long long)SPILL_CHUNK);
+          nb = atomicAdd(spill_counter, (unsigned long long)grab);
         wchunk = (long)__shfl(nb, 0, WAVE);
-        wleft = SPILL_CHUNK;
+        wleft = grab;
       }
       long o = wchunk + (long)(incl - my_ns);
       if (wave_total) {
@@ -1036,17 +1089,29 @@ __global__ __launch_bounds__(256) void tok_ablate_kernel(
 
 __global__ __launch_bounds__(256) void bucket_count_kernel(
     const u64* __restrict__ hashes, const u64* __restrict__ pos,
-    const i64* __restrict__ bucket_off,  // [nbuckets+1] exclusive offsets
+    const i64* __restrict__ bucket_off,  // [nbuckets+1] exclusive offsets,
+                                         // OR per-bucket lengths when
+                                         // region_stride > 0 (bucketed
+                                         // direct spill: bucket b lives at
+                                         // [b*stride, b*stride+len[b]))
     int nbuckets, int slices, u64* __restrict__ tkeys,
-    i64* __restrict__ tvals, u64* __restrict__ texm, u64 cap_mask) {
+    i64* __restrict__ tvals, u64* __restrict__ texm, u64 cap_mask,
+    long region_stride) {
   __shared__ u64 ckeys[BKT_SLOTS];
   __shared__ u64 cpos[BKT_SLOTS];
   __shared__ u32 ccnt[BKT_SLOTS];
   int bucket = blockIdx.x / slices;
   int slice = blockIdx.x % slices;
   if (bucket >= nbuckets) return;
-  long b0 = bucket_off[bucket];
-  long b1 = bucket_off[bucket + 1];
+  long b0, b1;
+  if (region_stride > 0) {
+    b0 = (long)bucket * region_stride;
+    long len = bucket_off[bucket];
+    b1 = b0 + (len < region_stride ? len : region_stride);
+  } else {
+    b0 = bucket_off[bucket];
+    b1 = bucket_off[bucket + 1];
+  }
   long bn = b1 - b0;
   long per = (bn + slices - 1) / slices;
   long s0 = b0 + (long)slice * per;

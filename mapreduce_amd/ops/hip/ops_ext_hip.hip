@@ -208,13 +208,31 @@ void tokenize_cache_spill(
       bool gpos = !(gp && gp[0] == '0');  // default ON: 4.90 vs 5.05 ms
       const char* ts = getenv("MR_TOK_TILE");
       int tsz = ts ? atoi(ts) : 4096;
+      // MR_SPILL_CHUNK ∈ {128, 512, 1024, 2048, 4096}: reservation-stall
+      // amortization vs HT_EMPTY pad waste.  Measured (Europarl shape,
+      // ~15M real misses over 8192 waves): 128=3.86, 512=1.93, 1024=1.92,
+      // 2048=1.84, 4096=3.09 ms/step — one grab per wave (2048) is the
+      // sweet spot; 4096 doubles reserved entries (19M pads flood the
+      // radix+bucket pipeline at ~75 ns each)
+      const char* sc = getenv("MR_SPILL_CHUNK");
+      int schunk = sc ? atoi(sc) : 2048;
       auto kfn = tokenize_v6_kernel<2048, false, 4096>;
       if (gpos && tsz == 8192) kfn = tokenize_v6_kernel<2048, true, 8192>;
       else if (gpos && cache == 4096) kfn = tokenize_v6_kernel<4096, true, 4096>;
       else if (gpos && cache == 1024) kfn = tokenize_v6_kernel<1024, true, 4096>;
+      else if (gpos && schunk == 128)
+        kfn = tokenize_v6_kernel<2048, true, 4096, 0, false, false, 128>;
+      else if (gpos && schunk == 1024)
+        kfn = tokenize_v6_kernel<2048, true, 4096, 0, false, false, 1024>;
+      else if (gpos && schunk == 2048)
+        kfn = tokenize_v6_kernel<2048, true, 4096, 0, false, false, 2048>;
+      else if (gpos && schunk == 4096)
+        kfn = tokenize_v6_kernel<2048, true, 4096, 0, false, false, 4096>;
       else if (gpos) kfn = tokenize_v6_kernel<2048, true, 4096>;
       else if (cache == 512) kfn = tokenize_v6_kernel<512, false, 4096>;
       else if (cache == 1024) kfn = tokenize_v6_kernel<1024, false, 4096>;
+      else if (schunk == 128)
+        kfn = tokenize_v6_kernel<2048, false, 4096, 0, false, false, 128>;
       long blocks = grid_for(n, TOK_BYTES * (tsz / 4096));
       static torch::Tensor cpos_g;  // persistent side-buffer (GPOS only)
       u64* cpg = nullptr;
@@ -258,6 +276,55 @@ void tokenize_cache_spill(
                          spill_cap,
                          reinterpret_cast<unsigned long long*>(nwords.data_ptr<i64>()));
   }
+}
+
+// ----------------------------------------------------- K2+K5 bucketed spill
+// v6 with NBKT=256: misses land directly in per-top-byte-bucket regions of
+// the (caller-owned) spill arrays — no radix_pass(56) bucketize afterwards.
+// counters: i64[256] per-bucket lengths (zeroed by the caller per job);
+// spill_bcap: per-bucket region capacity (out arrays hold 256*spill_bcap).
+// Overflowing buckets run their counter past spill_bcap with writes dropped
+// — the host detects via counters.max() > spill_bcap and fails loudly.
+void tokenize_cache_spill_bucketed(
+    torch::Tensor text, long pos_base, torch::Tensor tkeys,
+    torch::Tensor tvals, torch::Tensor texm, long spill_bcap,
+    torch::Tensor nwords, torch::Tensor out_hash, torch::Tensor out_pos,
+    torch::Tensor counters) {
+  TORCH_CHECK(text.is_cuda() && text.scalar_type() == torch::kUInt8 &&
+              text.is_contiguous(), "text must be contiguous u8 on GPU");
+  long n = text.numel();
+  long cap = tkeys.numel();
+  TORCH_CHECK((cap & (cap - 1)) == 0, "table capacity must be a power of 2");
+  TORCH_CHECK(counters.numel() >= 256, "counters must hold 256 buckets");
+  TORCH_CHECK(out_hash.numel() >= 256 * spill_bcap &&
+              out_pos.numel() >= 256 * spill_bcap, "spill arrays too small");
+  if (!n) return;
+  const char* gp = getenv("MR_TOK_GPOS");
+  bool gpos = !(gp && gp[0] == '0');
+  auto kfn = tokenize_v6_kernel<2048, false, 4096, 0, false, false, 512, 256>;
+  if (gpos)
+    kfn = tokenize_v6_kernel<2048, true, 4096, 0, false, false, 512, 256>;
+  long blocks = grid_for(n, TOK_BYTES);
+  static torch::Tensor cpos_gb;  // persistent GPOS side-buffer
+  u64* cpg = nullptr;
+  if (gpos) {
+    long need = blocks * 2048;
+    if (!cpos_gb.defined() || cpos_gb.numel() < need ||
+        cpos_gb.device() != text.device())
+      cpos_gb = torch::empty({need},
+                             torch::TensorOptions().device(text.device())
+                                 .dtype(torch::kInt64));
+    cpg = u64p(cpos_gb);
+  }
+  hipLaunchKernelGGL(kfn, dim3(blocks), dim3(kBlock), 0, cur_stream(),
+                     text.data_ptr<u8>(), n, (u64)pos_base, u64p(tkeys),
+                     tvals.data_ptr<i64>(),
+                     texm.numel() ? u64p(texm) : nullptr, (u64)(cap - 1),
+                     u64p(out_hash), u64p(out_pos),
+                     reinterpret_cast<unsigned long long*>(counters.data_ptr<i64>()),
+                     spill_bcap,
+                     reinterpret_cast<unsigned long long*>(nwords.data_ptr<i64>()),
+                     cpg, (const i64*)nullptr, 0, 0L);
 }
 
 // v6-structure ablation: mode 1=stage+classify, 2=+hash, 3=+cache, 0=full
@@ -351,16 +418,21 @@ double tok_ablate(torch::Tensor text, long mode, long iters) {
 void bucket_count(torch::Tensor hashes, torch::Tensor pos,
                   torch::Tensor bucket_off, long nbuckets, long slices,
                   torch::Tensor tkeys, torch::Tensor tvals,
-                  torch::Tensor texm) {
+                  torch::Tensor texm, long region_stride) {
   check_dev_i64(hashes, "hashes");
   long cap = tkeys.numel();
   TORCH_CHECK((cap & (cap - 1)) == 0, "table capacity must be a power of 2");
-  TORCH_CHECK(bucket_off.numel() == nbuckets + 1, "bucket_off size");
+  // region_stride > 0: bucket_off holds per-bucket LENGTHS and bucket b's
+  // data lives at [b*stride, b*stride+len[b]) (bucketed direct spill)
+  TORCH_CHECK(bucket_off.numel() >= (region_stride > 0 ? nbuckets
+                                                       : nbuckets + 1),
+              "bucket_off size");
   hipLaunchKernelGGL(bucket_count_kernel, dim3(nbuckets * slices),
                      dim3(kBlock), 0, cur_stream(), u64cp(hashes),
                      u64cp(pos), bucket_off.data_ptr<i64>(), (int)nbuckets,
                      (int)slices, u64p(tkeys), tvals.data_ptr<i64>(),
-                     texm.numel() ? u64p(texm) : nullptr, (u64)(cap - 1));
+                     texm.numel() ? u64p(texm) : nullptr, (u64)(cap - 1),
+                     region_stride);
 }
 
 // ----------------------------------------------------------------------- K5a
@@ -608,6 +680,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "tokenize -> compact (hash,pos) arrays");
   m.def("tokenize_cache_spill", &tokenize_cache_spill,
         "tokenize; LDS cache counts the head, misses spill");
+  m.def("tokenize_cache_spill_bucketed", &tokenize_cache_spill_bucketed,
+        "tokenize; misses spill directly into per-top-byte-bucket regions");
   m.def("tok_ablate", &tok_ablate, "ablation timing (diagnosis)");
   m.def("tok6_ablate", &tok6_ablate, "v6 ablation timing");
   m.def("bucket_count", &bucket_count,

@@ -259,6 +259,28 @@ def test_streaming_wordcount_mode_vs_counter(dev):
     assert dict(res2.to_host()) == got
 
 
+def test_bucketed_vs_chunked_spill_equivalence(dev, monkeypatch):
+    """MR_TOK_BSPILL=1 (per-bucket direct spill, no radix bucketize) and
+    =0 (wave-chunked spill + radix_pass(56)) must both match the oracle
+    and each other exactly."""
+    from mapreduce_amd.gpu.wordcount import WordCountJob
+    rng = np.random.default_rng(31)
+    vocab = [f"tk{i}q".encode() for i in range(8000)]  # > cache: real spill
+    widx = rng.integers(0, len(vocab), size=400_000)
+    data = b" ".join(vocab[i] for i in widx.tolist()) + b"\n"
+    text = torch.frombuffer(bytearray(data), dtype=torch.uint8).to(dev)
+    exp = collections.Counter(vocab[i] for i in widx.tolist())
+    results = {}
+    for flag in ("0", "1"):
+        monkeypatch.setenv("MR_TOK_BSPILL", flag)
+        job = WordCountJob(dev, vocab_estimate=16000, mode="streaming")
+        res = job.run(text)
+        assert res.nwords == len(widx)
+        results[flag] = dict(res.to_host())
+        assert results[flag] == dict(exp), f"MR_TOK_BSPILL={flag}"
+    assert results["0"] == results["1"]
+
+
 def test_wordcount_deterministic_across_runs(dev):
     """Counts are exactly reproducible run-to-run despite nondeterministic
     kernel scheduling (aggregation is commutative; exemplars may differ in
