@@ -1,0 +1,163 @@
+"""Iterative k-means clustering as a MapReduce job.
+
+Exercises the same reference machinery as train_digits (the APRIL-ANN
+iterative pattern): per-iteration mapfn computes partial
+cluster statistics on its data shard, reducefn sums them (declared
+associative+commutative — the job.lua:104-106 fast-path flags), finalfn
+recomputes centroids and returns "loop" until done.  Centroids cross
+processes via persistent_table (the reference's GridFS-model-file pattern,
+common.lua:57-77).
+
+The per-shard dataset is FIXED across iterations (deterministic per-shard
+seed), so inertia is mathematically non-increasing — the test asserts it.
+mapfn is device-aware: with init_args {"device": "cuda"} the distance
+argmin and partial sums run on the GPU.
+
+init_args: {"shards": int, "k": int, "dims": int, "points": int (per
+shard), "iters": int, "seed": int, "cnn": optional coordinator token,
+"db": str, "device": "cpu"|"cuda"}.
+"""
+
+from __future__ import annotations
+
+import base64
+import io
+import threading
+
+import torch
+
+_CFG = {"shards": 4, "k": 8, "dims": 16, "points": 5000, "iters": 5,
+        "seed": 7, "cnn": None, "db": "mr", "device": "cpu"}
+STATE = {"centroids": None, "iteration": 0, "inertia": [], "pt": None}
+
+associative_reducer = True
+commutative_reducer = True
+idempotent_reducer = True
+
+_MAP_LOCK = threading.Lock()  # threaded local mode shares STATE
+
+
+def _pt():
+    if STATE["pt"] is None and _CFG["cnn"]:
+        from mapreduce_amd.persistent_table import PersistentTable
+
+        STATE["pt"] = PersistentTable("kmeans", _CFG["cnn"], _CFG["db"])
+    return STATE["pt"]
+
+
+def _serialize(t: torch.Tensor) -> str:
+    buf = io.BytesIO()
+    torch.save(t.cpu(), buf)
+    return base64.b64encode(buf.getvalue()).decode()
+
+
+def _deserialize(blob: str) -> torch.Tensor:
+    return torch.load(io.BytesIO(base64.b64decode(blob)),
+                      weights_only=True, map_location="cpu")
+
+
+def _sync_from_pt():
+    pt = _pt()
+    if pt is None:
+        return
+    pt.update()
+    it = pt.get("iteration", 0)
+    blob = pt.get("centroids")
+    if blob is not None and it != STATE["iteration"]:
+        STATE["centroids"] = _deserialize(blob).to(_CFG["device"])
+        STATE["iteration"] = it
+
+
+def true_centers() -> torch.Tensor:
+    g = torch.Generator().manual_seed(_CFG["seed"])
+    return torch.randn(_CFG["k"], _CFG["dims"], generator=g) * 5.0
+
+
+def _shard_points(shard: int) -> torch.Tensor:
+    """Fixed per-shard blob data: points scattered around true centers."""
+    g = torch.Generator().manual_seed(10_000 + shard * 131 + _CFG["seed"])
+    n = _CFG["points"]
+    centers = true_centers()
+    which = torch.randint(0, _CFG["k"], (n,), generator=g)
+    pts = centers[which] + torch.randn(n, _CFG["dims"], generator=g)
+    return pts.to(_CFG["device"])
+
+
+def init(arg):
+    if arg:
+        _CFG.update({k: v for k, v in arg.items() if k in _CFG})
+    if STATE["centroids"] is None:
+        # deterministic k-means++ seeding over shard 0 (Forgy picks can
+        # merge blobs into one basin); greedy farthest-point variant
+        pts = _shard_points(0)
+        c = pts[:1].clone()
+        for _ in range(_CFG["k"] - 1):
+            d2 = torch.cdist(pts, c).min(dim=1).values
+            c = torch.cat([c, pts[int(d2.argmax())][None]])
+        STATE["centroids"] = c
+        STATE["iteration"] = 0
+        STATE["inertia"] = []
+
+
+def taskfn(emit):
+    for s in range(_CFG["shards"]):
+        emit(s + 1, {"shard": s})
+
+
+def mapfn(key, value, emit):
+    """Assign this shard's points to the nearest centroid; emit per-cluster
+    (sum_vector, count) partials and the shard's inertia contribution."""
+    with _MAP_LOCK:
+        _sync_from_pt()
+        c = STATE["centroids"]
+        pts = _shard_points(value["shard"])
+        d = torch.cdist(pts, c)  # [n, k]
+        mind, assign = d.min(dim=1)
+        out = []
+        for j in range(_CFG["k"]):
+            mask = assign == j
+            cnt = int(mask.sum())
+            if cnt:
+                s = pts[mask].sum(dim=0).cpu()
+                out.append((j, torch.cat([s, torch.tensor([float(cnt)])])))
+        inertia = float((mind ** 2).sum())
+    for j, partial in out:
+        emit(j, partial)
+    emit("__inertia__", torch.tensor([inertia]))
+
+
+def partitionfn(key):
+    from mapreduce_amd.utils.tuple import fnv1a32
+    return fnv1a32(key) % 4
+
+
+def reducefn(key, values, emit):
+    acc = values[0].clone()
+    for v in values[1:]:
+        acc += v
+    emit(acc)
+
+
+combinerfn = reducefn
+
+
+def finalfn(pairs):
+    c = STATE["centroids"].clone()
+    inertia = 0.0
+    for key, vals in pairs:
+        if key == "__inertia__":
+            inertia = float(vals[0].sum())
+        else:
+            v = vals[0]
+            cnt = float(v[-1])
+            if cnt > 0:  # empty cluster keeps its old centroid
+                c[key] = (v[:-1] / cnt).to(c.device)
+    STATE["centroids"] = c
+    STATE["inertia"].append(inertia)
+    STATE["iteration"] += 1
+    pt = _pt()
+    if pt is not None:
+        pt.set("centroids", _serialize(c))
+        pt.set("iteration", STATE["iteration"])
+        pt.update()
+    return "loop" if STATE["iteration"] < _CFG["iters"] else True
