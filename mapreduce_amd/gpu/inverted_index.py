@@ -53,6 +53,35 @@ class InvertedIndexResult:
                               tf[offs[i]:offs[i + 1]]))
         return out
 
+    def _signed_keys(self) -> torch.Tensor:
+        """keys hold u64 bit patterns; XOR the sign bit maps unsigned
+        order onto int64 order so torch.searchsorted works (cached)."""
+        sk = getattr(self, "_sk", None)
+        if sk is None:
+            sk = self.keys ^ (-1 << 63)
+            self._sk = sk
+        return sk
+
+    def lookup(self, word) -> list:
+        """Serve one word's postings [(doc, tf), ...] — O(log n) binary
+        search on the hash-sorted index, no host materialization."""
+        from mapreduce_amd.utils.tuple import wordhash64
+
+        if isinstance(word, str):
+            word = word.encode()
+        k = wordhash64(word)
+        ki = k - (1 << 64) if k >= (1 << 63) else k
+        sk = self._signed_keys()
+        q = torch.tensor([ki ^ (-1 << 63)], dtype=torch.int64,
+                         device=sk.device)
+        i = int(torch.searchsorted(sk, q).item())
+        if i >= self.keys.numel() or int(self.keys[i].item()) != ki:
+            return []
+        o0 = int(self.doc_offsets[i].item())
+        o1 = int(self.doc_offsets[i + 1].item())
+        return list(zip(self.docs[o0:o1].cpu().tolist(),
+                        self.tf[o0:o1].cpu().tolist()))
+
 
 _SM64_C0 = 0x9E3779B97F4A7C15 - (1 << 64)
 _SM64_C1 = 0xBF58476D1CE4E5B9 - (1 << 64)

@@ -58,4 +58,9 @@ def test_inverted_index_gpu(dev):
     text = torch.from_numpy(np.frombuffer(blob, dtype=np.uint8).copy()).to(dev)
     job = InvertedIndexJob(dev)
     res = job.run(text, splits)
-    assert res.to_host() == py_inverted_index(docs)
+    exp = py_inverted_index(docs)
+    assert res.to_host() == exp
+    # serving API: per-word postings lookup without materialization
+    for w in (vocab[0], vocab[123], vocab[499]):
+        assert res.lookup(w) == exp[w]
+    assert res.lookup(b"no-such-word-xyz") == []

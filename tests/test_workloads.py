@@ -59,6 +59,10 @@ def test_inverted_index_single_rank_cpu():
     got = res.to_host()
     exp = py_inverted_index(docs)
     assert got == exp
+    # lookup() serves single-word postings off the sorted index
+    for w in (vocab[0], vocab[57], vocab[199]):
+        assert res.lookup(w) == exp[w]
+    assert res.lookup("absent-word") == []
 
 
 def _free_port():
