@@ -37,6 +37,9 @@ class InvertedIndexResult:
     tf: torch.Tensor          # i64: term frequency of (word, doc)
     pos: torch.Tensor         # exemplar (off<<16|len) into blob_src
     blob_src: torch.Tensor    # u8
+    hash_kind: str = "wordhash64"  # GPU tier: wordhash64 (ops/hip
+                                   # common.h); CPU test tier: fnv1a64
+                                   # (ops/_cpu.py tokenize_words)
 
     def to_host(self):
         lens, blob = ops.extract_words(self.blob_src, self.pos)
@@ -65,11 +68,15 @@ class InvertedIndexResult:
     def lookup(self, word) -> list:
         """Serve one word's postings [(doc, tf), ...] — O(log n) binary
         search on the hash-sorted index, no host materialization."""
-        from mapreduce_amd.utils.tuple import wordhash64
-
         if isinstance(word, str):
             word = word.encode()
-        k = wordhash64(word)
+        if self.hash_kind == "wordhash64":
+            from mapreduce_amd.utils.tuple import wordhash64
+            k = wordhash64(word)
+        else:  # fnv1a64 (CPU test tier, ops/_cpu.py tokenize_words)
+            k = 0xCBF29CE484222325
+            for b in word:
+                k = ((k ^ b) * 0x100000001B3) & 0xFFFFFFFFFFFFFFFF
         ki = k - (1 << 64) if k >= (1 << 63) else k
         sk = self._signed_keys()
         q = torch.tensor([ki ^ (-1 << 63)], dtype=torch.int64,
@@ -293,5 +300,7 @@ class InvertedIndexJob:
             wk = uh
             wpos = up
             offsets = torch.zeros(1, dtype=torch.int64, device=dev)
-        return InvertedIndexResult(keys=wk, doc_offsets=offsets, docs=ud,
-                                   tf=tf, pos=wpos, blob_src=blob_src)
+        return InvertedIndexResult(
+            keys=wk, doc_offsets=offsets, docs=ud, tf=tf, pos=wpos,
+            blob_src=blob_src,
+            hash_kind="wordhash64" if dev.type == "cuda" else "fnv1a64")
