@@ -41,6 +41,8 @@ def main() -> int:
                    choices=["auto", "streaming", "fused"])
     p.add_argument("--uncoordinated", action="store_true",
                    help="bypass the control plane (engine-only timing)")
+    p.add_argument("--timing", action="store_true",
+                   help="print per-phase HIP-event times to stderr")
     args = p.parse_args()
 
     from mapreduce_amd import ops
@@ -55,7 +57,7 @@ def main() -> int:
     corpus = make_corpus(device, nwords=args.words, nsplits=args.splits,
                          vocab_size=args.vocab, seed=1234 + rank)
     job = WordCountJob(device, vocab_estimate=max(args.vocab, 1 << 12),
-                       mode=args.mode)
+                       mode=args.mode, timing=args.timing)
     splits = corpus.splits()
     runner = None
     if not args.uncoordinated:
@@ -109,6 +111,11 @@ def main() -> int:
     value = total_words_per_step * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
+    if args.timing and job.last_phase_ms:
+        print(f"[rank {rank}] phase_ms: "
+              + " ".join(f"{k}={v:.3f}" for k, v in
+                         job.last_phase_ms.items()),
+              file=sys.stderr, flush=True)
     if rank == 0:
         out = {
             "metric": "words/sec",

@@ -172,16 +172,24 @@ class GpuClusterRunner:
         return result
 
     def job_stats(self) -> dict:
-        """Scan this rank's job docs (observability parity, C9)."""
+        """Scan this rank's job docs (observability parity, C9).  When the
+        engine was built with timing=True, per-phase HIP-event times ride
+        along — the reference's cpu/real-time stats sub-document
+        (server.lua:584-601) in kernel-time form."""
         ns = self._ns()
+        extra = {}
+        phase_ms = getattr(self.job, "last_phase_ms", None)
+        if phase_ms:
+            extra["phase_ms"] = {k: round(v, 4) for k, v in phase_ms.items()}
         if self.claim_mode == "batch":
             doc, _ = self.coord.get_doc(f"{ns}/batch")
             return {"jobs": len(doc["splits"]) if doc else 0,
-                    "status": doc["status"] if doc else None}
+                    "status": doc["status"] if doc else None, **extra}
         docs = [self.coord.get_doc(f"{ns}/{i}")[0]
                 for i in self.coord.get_ids(ns)]
         return {
             "jobs": len(docs),
             "written": sum(d["status"] == STATUS.WRITTEN for d in docs),
             "broken": sum(d["status"] == STATUS.BROKEN for d in docs),
+            **extra,
         }
