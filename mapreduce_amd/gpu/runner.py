@@ -193,3 +193,33 @@ class GpuClusterRunner:
             "broken": sum(d["status"] == STATUS.BROKEN for d in docs),
             **extra,
         }
+
+    def cluster_stats(self) -> dict:
+        """C9: reduction of per-rank timing vectors as a collective — the
+        reference aggregated job-doc timestamps with Mongo server-side JS
+        map-reduce (server.lua:155-183, :540-555); here one allreduce
+        carries every rank's phase times.  Phase names must match across
+        ranks (they do: all ranks run the same phase sequence).  Returns
+        {phase: {"max": ms, "mean": ms}} — max is the wall-clock bound
+        (slowest rank), mean shows skew."""
+        import torch
+
+        pm = getattr(self.job, "last_phase_ms", None) or {}
+        names = sorted(pm)
+        if not names:
+            return {}
+        world = getattr(self.job, "world", 1)
+        vals = torch.tensor([float(pm[n]) for n in names],
+                            dtype=torch.float64)
+        mx, sm = vals, vals
+        if world > 1:
+            import torch.distributed as td
+            if td.is_available() and td.is_initialized():
+                if td.get_backend(self.group) == "nccl":
+                    vals = vals.cuda()
+                mx = vals.clone()
+                td.all_reduce(mx, op=td.ReduceOp.MAX, group=self.group)
+                sm = vals.clone()
+                td.all_reduce(sm, op=td.ReduceOp.SUM, group=self.group)
+        return {n: {"max": float(mx[i]), "mean": float(sm[i]) / world}
+                for i, n in enumerate(names)}

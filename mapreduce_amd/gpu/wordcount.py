@@ -108,24 +108,33 @@ class WordCountJob:
         self.table = ops.make_table(vocab_estimate, self.device)
         self._nwords = torch.zeros(1, dtype=torch.int64, device=self.device)
         # per-phase tracing (the job-document timestamps of job.lua:117-152
-        # in HIP-event form; stats format parity with server.lua:557-602)
-        self.timing = timing and self.device.type == "cuda"
+        # in HIP-event form; stats format parity with server.lua:557-602);
+        # on CPU devices wall-clock marks keep the same stats shape
+        self.timing = timing
+        self._hip_events = timing and self.device.type == "cuda"
         self.last_phase_ms: dict = {}
         self._events: list = []
 
     def _mark(self, name: str) -> None:
-        if self.timing:
+        if not self.timing:
+            return
+        if self._hip_events:
             ev = torch.cuda.Event(enable_timing=True)
             ev.record()
-            self._events.append((name, ev))
+        else:
+            import time
+            ev = time.perf_counter()
+        self._events.append((name, ev))
 
     def _collect_timing(self) -> None:
         if not self.timing or len(self._events) < 2:
             return
-        self._events[-1][1].synchronize()
+        if self._hip_events:
+            self._events[-1][1].synchronize()
         out = {}
         for (n0, e0), (n1, e1) in zip(self._events, self._events[1:]):
-            out[n1] = e0.elapsed_time(e1)
+            out[n1] = (e0.elapsed_time(e1) if self._hip_events
+                       else (e1 - e0) * 1000.0)
         self.last_phase_ms = out
         self._events = []
 
@@ -142,6 +151,8 @@ class WordCountJob:
     def begin_map(self, text: torch.Tensor) -> None:
         self.reset()
         self._text = text
+        self._events = []
+        self._mark("start")
         if self.mode == "streaming":
             import os
             opts = dict(dtype=torch.int64, device=self.device)
@@ -203,6 +214,12 @@ class WordCountJob:
     def finish_map(self) -> int:
         """Drain the spill through bucketize + per-bucket LDS count.
         Returns the word count (the one host sync of the map phase)."""
+        self._mark("map_tokenize")
+        n = self._finish_map_inner()
+        self._mark("map_combine")
+        return n
+
+    def _finish_map_inner(self) -> int:
         if self.mode == "streaming":
             import os
             slices = int(os.environ.get("MR_BUCKET_SLICES", "32"))
@@ -272,20 +289,16 @@ class WordCountJob:
         # launches serialize (measured 98% of step time before fusing).
         # Split boundaries are whitespace-aligned, so tokenization over the
         # coalesced range is byte-identical to per-split runs.
+        # phase marks live inside the phase methods so the cluster
+        # runner's claim-driven execution traces identically
         self.begin_map(text)
-        self._mark("start")
         if self._coalesced(text, splits):
             self.map_split(splits[0][0], splits[-1][1])
         else:
             for (s, e) in splits:
                 self.map_split(s, e)
-        self._mark("map_tokenize")
         nwords = self.finish_map()
-        self._mark("map_combine")
-        res = self.shuffle_reduce(nwords)
-        self._mark("shuffle_reduce")
-        self._collect_timing()
-        return res
+        return self.shuffle_reduce(nwords)
 
     def shuffle_reduce(self, nwords: int) -> WordCountResult:
         """Phase 2: extract + sort uniques, all-to-all exchange, segmented
@@ -344,5 +357,7 @@ class WordCountJob:
             fk, fv, fp = sk, sv, sp
             blob_src = text
 
+        self._mark("shuffle_reduce")
+        self._collect_timing()
         return WordCountResult(keys=fk, counts=fv, pos=fp,
                                blob_src=blob_src, nwords=nwords)

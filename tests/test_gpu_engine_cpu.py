@@ -81,3 +81,43 @@ def test_wordcount_job_gloo_ws2(tmp_path):
     s.close()
     torch.multiprocessing.spawn(
         _dist_worker, args=(2, port, str(tmp_path)), nprocs=2, join=True)
+
+
+def _timing_worker(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.distributed.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from mapreduce_amd.gpu.runner import GpuClusterRunner
+
+        c = make_corpus("cpu", nwords=5_000, nsplits=3, vocab_size=300,
+                        seed=50 + rank)
+        job = WordCountJob("cpu", vocab_estimate=600, timing=True)
+        runner = GpuClusterRunner(job, claim_mode="batch")
+        runner.run(c.text, c.splits())
+        # C9: per-rank phase times reduced across ranks as a collective
+        cs = runner.cluster_stats()
+        assert set(cs) >= {"map_tokenize", "shuffle_reduce"}
+        for ph, d in cs.items():
+            assert d["max"] > 0 and d["max"] >= d["mean"] - 1e-9, (ph, d)
+        # both ranks see identical reduced values
+        all_cs = [None] * world
+        torch.distributed.all_gather_object(all_cs, cs)
+        if rank == 0:
+            assert all(
+                abs(all_cs[0][p]["max"] - all_cs[1][p]["max"]) < 1e-9
+                for p in cs)
+        # per-rank stats carry the raw phase_ms
+        assert "phase_ms" in runner.job_stats()
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_cluster_timing_stats_gloo_ws2():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    torch.multiprocessing.spawn(_timing_worker, args=(2, port), nprocs=2,
+                                join=True)
