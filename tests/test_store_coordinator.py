@@ -98,3 +98,34 @@ def test_drop_ns(pair):
         time.sleep(0.05)
     assert m.get_ids("map_jobs") == []
     assert m.get_doc("map_jobs/1")[0] is None
+
+
+@pytest.mark.timeout(300)
+def test_bulk_docs_and_error_flood(pair):
+    """The reference unit-tested cnn's 50k batched insert + error channel
+    (cnn.lua:126-168).  Equivalent here: thousands of job docs and error
+    records through one TCPStore without loss or cross-talk."""
+    master, client = pair
+    n = 5000
+    for i in range(n):
+        client.set_doc(f"bulk/{i}", {"i": i, "payload": "x" * 50})
+    client.set_ids("bulk", [str(i) for i in range(n)])
+    ids = master.get_ids("bulk")
+    assert len(ids) == n
+    # spot-check + full count via sampled reads
+    for i in (0, 1, n // 2, n - 1):
+        doc, _ = master.get_doc(f"bulk/{i}")
+        assert doc == {"i": i, "payload": "x" * 50}
+    # error flood from two producers, drained in order of arrival count
+    for i in range(200):
+        (client if i % 2 else master).insert_error(f"w{i % 2}", f"err {i}")
+    drained = 0
+    seen = []
+    while drained < 200:
+        errs, drained = master.get_errors(drained)
+        seen.extend(errs)
+    assert len(seen) == 200
+    assert {e["who"] for e in seen} == {"w0", "w1"}
+    assert sorted(int(e["msg"].split()[1]) for e in seen) == list(range(200))
+    master.drop_ns("bulk")
+    assert master.get_ids("bulk") == []
