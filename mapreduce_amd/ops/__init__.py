@@ -141,12 +141,28 @@ def make_table(capacity: int, device, exemplar: bool = True):
     return CpuHashTable(capacity, device, exemplar)
 
 
+_SMALL_SORT_N = 1 << 20  # below this, one stable torch.sort dispatch
+                         # (rocPRIM segmented radix) beats 8 host-driven
+                         # hist+scan+scatter passes on launch overhead
+
+
 def sort_pairs(keys: torch.Tensor, vals: Optional[torch.Tensor] = None,
                bits: int = 64):
     """Stable LSD radix sort of u64 keys (bit-pattern order) w/ payload."""
     if not keys.is_cuda:
         from . import _cpu
         return _cpu.sort_pairs(keys, vals, bits)
+    import os
+    thresh = int(os.environ.get("MR_SMALL_SORT_N", _SMALL_SORT_N))
+    if keys.numel() < thresh:
+        if os.environ.get("MR_SMALL_SORT_TORCH", "1") == "1":
+            # unsigned order via sign-bit flip; stable to preserve the
+            # LSD composite-sort contract (inverted index doc pass)
+            sk = keys ^ (-1 << 63)
+            _, perm = torch.sort(sk, stable=True)
+            k = keys.index_select(0, perm)
+            return (k, vals.index_select(0, perm)
+                    if vals is not None else None)
     empty = torch.empty(0, dtype=torch.int64, device=keys.device)
     k, v = ext().radix_sort_pairs(keys.contiguous(),
                                   vals.contiguous() if vals is not None
