@@ -194,8 +194,12 @@ class InvertedIndexJob:
             bucket_off = torch.zeros(257, dtype=torch.int64, device=dev)
             torch.cumsum(totals, 0, out=bucket_off[1:])
             table = ops.make_table(max(1 << 16, k2.numel() // 6), dev)
+            # 2048 LDS slots: composite (word,doc) keys run ~3-5k distinct
+            # per slice — the smaller wordcount table would overflow into
+            # the per-element ht_add fallback
             ops.ext().bucket_count(hk, pv, bucket_off, 256, 32,
-                                   table.tkeys, table.tvals, table.texm, 0)
+                                   table.tkeys, table.tvals, table.texm,
+                                   0, 2048)
             uk2, tf, upos = table.extract()
             ud = torch.searchsorted(starts, upos >> 16, right=True) - 1
             ud = ud + self.doc_base

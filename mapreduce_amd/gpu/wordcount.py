@@ -237,7 +237,7 @@ class WordCountJob:
                     ops.ext().bucket_count(
                         self._spill_h, self._spill_p, self._spill_c, 256,
                         slices, self.table.tkeys, self.table.tvals,
-                        self.table.texm, self._spill_bcap)
+                        self.table.texm, self._spill_bcap, 1024)
                 self._spill_h = self._spill_p = None
                 return n
             nspill = int(self._spill_c.item())
@@ -260,9 +260,13 @@ class WordCountJob:
                 # in bucket 255 (top byte 0xFF), so fine slicing keeps its
                 # blocks off the critical path (sweep: 8=3.19, 16=2.58,
                 # 32=2.38, 64=2.40 ms/step)
+                # 1024 LDS slots: ~390 distinct/slice fits 2.6x over;
+                # 20 KB = 8 blocks/CU (sweep: 2048=1.71, 1024=1.60,
+                # 512=1.64 ms/step — the kernel is occupancy/latency
+                # bound, profiles/pmc_final_kernels.txt)
                 ops.ext().bucket_count(hk, pv, bucket_off, 256, slices,
                                        self.table.tkeys, self.table.tvals,
-                                       self.table.texm, 0)
+                                       self.table.texm, 0, 1024)
             self._spill_h = self._spill_p = None
             return n
         return int(self._nwords.item())

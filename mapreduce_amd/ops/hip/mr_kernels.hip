@@ -1085,8 +1085,14 @@ __global__ __launch_bounds__(256) void tok_ablate_kernel(
 // the global table sees only per-block flushes of distinct keys.
 // ---------------------------------------------------------------------------
 
-#define BKT_SLOTS 2048  // LDS table slots (power of 2)
+#define BKT_SLOTS 2048  // default LDS table slots (power of 2)
 
+// SLOTS trades LDS table capacity against occupancy: 2048 slots = 40 KB
+// = 4 blocks/CU; 1024 = 20 KB = 8 blocks/CU (PMC: waves wait ~16k cycles
+// per ~660 VALU instrs — latency-bound, occupancy is the lever).  A slice
+// holds ~390 distinct keys on the Europarl shape; overflow degrades to
+// per-element ht_add, never wrong.
+template <int SLOTS>
 __global__ __launch_bounds__(256) void bucket_count_kernel(
     const u64* __restrict__ hashes, const u64* __restrict__ pos,
     const i64* __restrict__ bucket_off,  // [nbuckets+1] exclusive offsets,
@@ -1097,9 +1103,9 @@ __global__ __launch_bounds__(256) void bucket_count_kernel(
     int nbuckets, int slices, u64* __restrict__ tkeys,
     i64* __restrict__ tvals, u64* __restrict__ texm, u64 cap_mask,
     long region_stride) {
-  __shared__ u64 ckeys[BKT_SLOTS];
-  __shared__ u64 cpos[BKT_SLOTS];
-  __shared__ u32 ccnt[BKT_SLOTS];
+  __shared__ u64 ckeys[SLOTS];
+  __shared__ u64 cpos[SLOTS];
+  __shared__ u32 ccnt[SLOTS];
   int bucket = blockIdx.x / slices;
   int slice = blockIdx.x % slices;
   if (bucket >= nbuckets) return;
@@ -1117,7 +1123,7 @@ __global__ __launch_bounds__(256) void bucket_count_kernel(
   long s0 = b0 + (long)slice * per;
   long s1 = s0 + per < b1 ? s0 + per : b1;
   if (s0 >= s1) return;
-  for (int s = threadIdx.x; s < BKT_SLOTS; s += blockDim.x) {
+  for (int s = threadIdx.x; s < SLOTS; s += blockDim.x) {
     ckeys[s] = HT_EMPTY;
     ccnt[s] = 0;
   }
@@ -1126,7 +1132,7 @@ __global__ __launch_bounds__(256) void bucket_count_kernel(
     u64 k = hashes[i];
     if (k == HT_EMPTY) continue;  // spill-chunk padding (tokenize_v6)
     u64 p = pos[i];
-    u32 slot = (u32)((k ^ (k >> 17)) & (BKT_SLOTS - 1));
+    u32 slot = (u32)((k ^ (k >> 17)) & (SLOTS - 1));
     bool done = false;
     for (int pr = 0; pr < 64; ++pr) {
       u64 cur = ckeys[slot];
@@ -1146,13 +1152,13 @@ __global__ __launch_bounds__(256) void bucket_count_kernel(
           break;
         }
       }
-      slot = (slot + 1) & (BKT_SLOTS - 1);
+      slot = (slot + 1) & (SLOTS - 1);
     }
     if (!done)  // pathological bucket: spill straight to the global table
       ht_add(k, p, 1, tkeys, tvals, texm, cap_mask);
   }
   __syncthreads();
-  for (int s = threadIdx.x; s < BKT_SLOTS; s += blockDim.x)
+  for (int s = threadIdx.x; s < SLOTS; s += blockDim.x)
     if (ckeys[s] != HT_EMPTY && ccnt[s])
       ht_add(ckeys[s], cpos[s], (i64)ccnt[s], tkeys, tvals, texm, cap_mask);
 }

@@ -421,7 +421,7 @@ double tok_ablate(torch::Tensor text, long mode, long iters) {
 void bucket_count(torch::Tensor hashes, torch::Tensor pos,
                   torch::Tensor bucket_off, long nbuckets, long slices,
                   torch::Tensor tkeys, torch::Tensor tvals,
-                  torch::Tensor texm, long region_stride) {
+                  torch::Tensor texm, long region_stride, long slots_arg) {
   check_dev_i64(hashes, "hashes");
   long cap = tkeys.numel();
   TORCH_CHECK((cap & (cap - 1)) == 0, "table capacity must be a power of 2");
@@ -430,7 +430,16 @@ void bucket_count(torch::Tensor hashes, torch::Tensor pos,
   TORCH_CHECK(bucket_off.numel() >= (region_stride > 0 ? nbuckets
                                                        : nbuckets + 1),
               "bucket_off size");
-  hipLaunchKernelGGL(bucket_count_kernel, dim3(nbuckets * slices),
+  // slots: caller picks by expected distinct-per-slice (wordcount ~390 ->
+  // 1024 = 8 blocks/CU; inverted index ~3-5k -> 2048); MR_BKT_SLOTS env
+  // overrides for A/B
+  const char* bs = getenv("MR_BKT_SLOTS");
+  int slots = bs ? atoi(bs) : (int)slots_arg;
+  if (slots != 512 && slots != 1024 && slots != 2048) slots = 2048;
+  auto kfn = bucket_count_kernel<2048>;
+  if (slots == 1024) kfn = bucket_count_kernel<1024>;
+  else if (slots == 512) kfn = bucket_count_kernel<512>;
+  hipLaunchKernelGGL(kfn, dim3(nbuckets * slices),
                      dim3(kBlock), 0, cur_stream(), u64cp(hashes),
                      u64cp(pos), bucket_off.data_ptr<i64>(), (int)nbuckets,
                      (int)slices, u64p(tkeys), tvals.data_ptr<i64>(),
