@@ -30,6 +30,28 @@ hipStream_t cur_stream() { return at::hip::getCurrentHIPStream(); }
 
 // LDS-staged scatter (v2) is the default — direct scatter (v1) measured
 // ~2.5x slower on write coalescing; MR_RADIX_V1=1 switches back for A/B.
+// MR_RS_ITEMS ∈ {4, 8}: radix tile = 256*ITEMS.  8 = longer digit runs
+// (better write coalescing), 4 = half the LDS stage (38->22 KB, 4->7
+// blocks/CU).  v1 (direct scatter) stays ITEMS=8.
+int rs_items() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("MR_RS_ITEMS");
+    const char* v1 = getenv("MR_RADIX_V1");  // v1 kernel is ITEMS=8 only
+    v = (e && atoi(e) == 4 && !(v1 && v1[0] == '1')) ? 4 : 8;
+  }
+  return v;
+}
+
+long rs_tile() { return (long)RS_BLOCK * rs_items(); }
+
+void launch_radix_hist(const u64* kin, long n, int shift, long ntiles,
+                       i64* hist) {
+  auto kfn = rs_items() == 4 ? radix_hist_kernel<4> : radix_hist_kernel<8>;
+  hipLaunchKernelGGL(kfn, dim3(ntiles), dim3(RS_BLOCK), 0, cur_stream(),
+                     kin, n, shift, ntiles, hist);
+}
+
 void launch_radix_scatter(const u64* kin, const u64* vin, long n, int shift,
                           long ntiles, const i64* base, u64* kout,
                           u64* vout) {
@@ -42,10 +64,13 @@ void launch_radix_scatter(const u64* kin, const u64* vin, long n, int shift,
     hipLaunchKernelGGL(radix_scatter_kernel, dim3(ntiles), dim3(RS_BLOCK), 0,
                        cur_stream(), kin, vin, n, shift, ntiles, base, kout,
                        vout);
-  else
-    hipLaunchKernelGGL(radix_scatter_v2_kernel, dim3(ntiles), dim3(RS_BLOCK),
+  else {
+    auto kfn = rs_items() == 4 ? radix_scatter_v2_kernel<4>
+                               : radix_scatter_v2_kernel<8>;
+    hipLaunchKernelGGL(kfn, dim3(ntiles), dim3(RS_BLOCK),
                        0, cur_stream(), kin, vin, n, shift, ntiles, base,
                        kout, vout);
+  }
 }
 
 u64* u64p(torch::Tensor& t) { return reinterpret_cast<u64*>(t.data_ptr<i64>()); }
@@ -630,15 +655,14 @@ std::vector<torch::Tensor> radix_pass(torch::Tensor keys, torch::Tensor vals,
   long n = keys.numel();
   bool has_vals = vals.numel() > 0;
   auto opts = keys.options();
-  long ntiles = (n + RS_TILE - 1) / RS_TILE;
+  long ntiles = (n + rs_tile() - 1) / rs_tile();
   if (ntiles == 0) ntiles = 1;
   auto kout = torch::empty({n}, opts);
   auto vout = has_vals ? torch::empty({n}, opts) : torch::empty({0}, opts);
   auto hist = torch::zeros({(long)RS_BINS * ntiles}, opts);
   if (n) {
-    hipLaunchKernelGGL(radix_hist_kernel, dim3(ntiles), dim3(RS_BLOCK), 0,
-                       cur_stream(), u64cp(keys), n, (int)shift, ntiles,
-                       hist.data_ptr<i64>());
+    launch_radix_hist(u64cp(keys), n, (int)shift, ntiles,
+                      hist.data_ptr<i64>());
     auto scanned = torch::cumsum(hist, 0);
     auto base = scanned - hist;
     launch_radix_scatter(u64cp(keys), has_vals ? u64cp(vals) : nullptr, n,
@@ -657,7 +681,7 @@ std::vector<torch::Tensor> radix_sort_pairs(torch::Tensor keys,
   if (n == 0) return {keys, vals};
   TORCH_CHECK(bits >= 1 && bits <= 64, "bits in [1,64]");
   int passes = (bits + 7) / 8;
-  long ntiles = (n + RS_TILE - 1) / RS_TILE;
+  long ntiles = (n + rs_tile() - 1) / rs_tile();
   auto opts = keys.options();
   auto kbuf = torch::empty({n}, opts);
   auto vbuf = has_vals ? torch::empty({n}, opts) : torch::empty({0}, opts);
@@ -666,9 +690,7 @@ std::vector<torch::Tensor> radix_sort_pairs(torch::Tensor keys,
   torch::Tensor kin = keys, vin = vals, kout = kbuf, vout = vbuf;
   for (int p = 0; p < passes; ++p) {
     int shift = p * 8;
-    hipLaunchKernelGGL(radix_hist_kernel, dim3(ntiles), dim3(RS_BLOCK), 0,
-                       cur_stream(), u64cp(kin), n, shift, ntiles,
-                       hist.data_ptr<i64>());
+    launch_radix_hist(u64cp(kin), n, shift, ntiles, hist.data_ptr<i64>());
     // exclusive scan over the digit-major flat histogram = base[d][t]
     auto scanned = torch::cumsum(hist, 0);
     auto base = scanned - hist;

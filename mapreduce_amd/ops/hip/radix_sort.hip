@@ -21,6 +21,11 @@
 #define RS_TILE (RS_BLOCK * RS_ITEMS)
 #define RS_BINS 256
 
+// ITEMS (elements/thread) sets the tile: 8 -> 2048 (38 KB LDS in the
+// v2 scatter, 4 blocks/CU); 4 -> 1024 (22 KB, 7 blocks/CU) — occupancy
+// vs digit-run length (write coalescing) tradeoff, selected per size by
+// MR_RS_ITEMS.
+template <int ITEMS>
 __global__ __launch_bounds__(RS_BLOCK) void radix_hist_kernel(
     const u64* __restrict__ keys, long n, int shift, long ntiles,
     i64* __restrict__ hist) {
@@ -28,8 +33,8 @@ __global__ __launch_bounds__(RS_BLOCK) void radix_hist_kernel(
   for (int b = threadIdx.x; b < RS_BINS; b += blockDim.x) lh[b] = 0;
   __syncthreads();
   long tile = blockIdx.x;
-  long base = tile * RS_TILE;
-  for (int r = 0; r < RS_ITEMS; ++r) {
+  long base = tile * (RS_BLOCK * ITEMS);
+  for (int r = 0; r < ITEMS; ++r) {
     long i = base + r * RS_BLOCK + threadIdx.x;
     if (i < n) {
       u32 d = (u32)((keys[i] >> shift) & 0xFF);
@@ -49,6 +54,7 @@ __global__ __launch_bounds__(RS_BLOCK) void radix_hist_kernel(
 // (avg run = TILE/256 elements), restoring write coalescing.
 // digit_start_in_tile comes from a 256-entry LDS prefix over this tile's
 // histogram (recomputed; must equal radix_hist_kernel's counts).
+template <int ITEMS>
 __global__ __launch_bounds__(RS_BLOCK) void radix_scatter_v2_kernel(
     const u64* __restrict__ keys, const u64* __restrict__ vals, long n,
     int shift, long ntiles, const i64* __restrict__ base_dx,
@@ -56,12 +62,12 @@ __global__ __launch_bounds__(RS_BLOCK) void radix_scatter_v2_kernel(
   __shared__ u32 wavecnt[RS_WAVES][RS_BINS];
   __shared__ u32 cnt_base[RS_BINS];     // running per-digit counts
   __shared__ u32 digit_start[RS_BINS + 1];
-  __shared__ u64 stage_k[RS_TILE];
-  __shared__ u64 stage_v[RS_TILE];
+  __shared__ u64 stage_k[(RS_BLOCK * ITEMS)];
+  __shared__ u64 stage_v[(RS_BLOCK * ITEMS)];
   for (int b = threadIdx.x; b < RS_BINS; b += blockDim.x) cnt_base[b] = 0;
   long tile = blockIdx.x;
-  long tbase = tile * RS_TILE;
-  long tile_n = n - tbase < RS_TILE ? n - tbase : RS_TILE;
+  long tbase = tile * (RS_BLOCK * ITEMS);
+  long tile_n = n - tbase < (RS_BLOCK * ITEMS) ? n - tbase : (RS_BLOCK * ITEMS);
   int wave = threadIdx.x / WAVE;
   int lane = threadIdx.x % WAVE;
   u64 lt_mask = ((u64)1 << lane) - 1;
@@ -69,7 +75,7 @@ __global__ __launch_bounds__(RS_BLOCK) void radix_scatter_v2_kernel(
 
   // ---- per-tile digit histogram + exclusive prefix -> digit_start
   __syncthreads();
-  for (int r = 0; r < RS_ITEMS; ++r) {
+  for (int r = 0; r < ITEMS; ++r) {
     long i = tbase + (long)r * RS_BLOCK + threadIdx.x;
     if (i < n)
       atomicAdd(&cnt_base[(u32)((keys[i] >> shift) & 0xFF)], 1u);
@@ -88,7 +94,7 @@ __global__ __launch_bounds__(RS_BLOCK) void radix_scatter_v2_kernel(
   __syncthreads();
 
   // ---- stable rank + stage into LDS at (digit_start + tile_rank)
-  for (int r = 0; r < RS_ITEMS; ++r) {
+  for (int r = 0; r < ITEMS; ++r) {
     for (int b = threadIdx.x; b < RS_BINS; b += blockDim.x)
       for (int w = 0; w < RS_WAVES; ++w) wavecnt[w][b] = 0;
     __syncthreads();
