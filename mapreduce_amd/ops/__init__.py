@@ -112,7 +112,9 @@ class HashTable:
         shared-counter atomic serializes cross-XCD — measured 6.3 ms on a
         2^25-slot scan) and mask out its HT_EMPTY chunk padding here;
         small tables keep the padding-free path."""
-        if self.cap >= (1 << 22):
+        import os
+        v2_min = int(os.environ.get("MR_EXTRACT_V2_MIN", 1 << 22))
+        if self.cap >= v2_min:
             k, v, p, c = ext().hash_extract_v2(self.tkeys, self.tvals,
                                                self.texm)
             n = int(c.item())
