@@ -662,14 +662,18 @@ template <int CACHE_N, bool GPOS, int TILE_N, int MODE = 0,
           int SCHUNK = 512,  // chunked-allocator reservation size: pads
                              // scale with it (waves x SCHUNK/2 wasted
                              // tail entries) vs atomic amortization
-          int NBKT = 0>  // >0: bucketed direct spill — misses go straight
+          int NBKT = 0,  // >0: bucketed direct spill — misses go straight
                          // into per-top-byte-bucket regions (out[b*cap ..])
                          // via wave-cooperative reservation on NBKT
                          // counters, replacing the single-array spill +
                          // the later radix_pass(56) bucketize entirely
                          // (that pass measured ~320 us/step: hist 63 +
                          // scatter 255, profiles/kernel_stats_final_step)
-__global__ __launch_bounds__(256) void tokenize_v6_kernel(
+          int BLOCKN = 256>  // threads/block; one window pass covers
+                             // BLOCKN*16 bytes, so TILE_N scales with it
+                             // (512/8192 trades +2 KB tile LDS for 2x
+                             // waves per block — occupancy probe)
+__global__ __launch_bounds__(BLOCKN) void tokenize_v6_kernel(
     const u8* __restrict__ text, long n, u64 pos_base,
     u64* __restrict__ tkeys, i64* __restrict__ tvals, u64* __restrict__ texm,
     u64 cap_mask, u64* __restrict__ out_hash, u64* __restrict__ out_pos,
@@ -715,8 +719,8 @@ __global__ __launch_bounds__(256) void tokenize_v6_kernel(
       }
     }
     __syncthreads();
-    for (int wnd = 0; wnd < TILE_N / 4096; ++wnd) {
-    long my0 = ((long)tid + (long)wnd * 256) * TOK_BYTES;
+    for (int wnd = 0; wnd < TILE_N / (BLOCKN * TOK_BYTES); ++wnd) {
+    long my0 = ((long)tid + (long)wnd * BLOCKN) * TOK_BYTES;
     // whole wave stays in the loop (the spill wave-scan below needs every
     // lane present); out-of-range lanes just contribute ns = 0
     bool wactive = my0 < avail;

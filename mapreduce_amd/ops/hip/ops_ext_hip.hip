@@ -245,8 +245,17 @@ void tokenize_cache_spill(
       // radix+bucket pipeline at ~75 ns each)
       const char* sc = getenv("MR_SPILL_CHUNK");
       int schunk = sc ? atoi(sc) : 2048;
+      // MR_TOK_BLOCK=512: 8 KB tile in ONE window pass (same per-thread
+      // geometry as 256/4096) — +2 KB LDS for 2x waves/block, occupancy
+      // probe for the 68%-wait tokenizer
+      const char* tb = getenv("MR_TOK_BLOCK");
+      int blockn = (tb && atoi(tb) == 512) ? 512 : 256;
+      if (blockn == 512) tsz = 8192;
       auto kfn = tokenize_v6_kernel<2048, false, 4096>;
-      if (gpos && tsz == 8192) kfn = tokenize_v6_kernel<2048, true, 8192>;
+      if (gpos && blockn == 512)
+        kfn = tokenize_v6_kernel<2048, true, 8192, 0, false, false, 2048,
+                                 0, 512>;
+      else if (gpos && tsz == 8192) kfn = tokenize_v6_kernel<2048, true, 8192>;
       else if (gpos && cache == 4096) kfn = tokenize_v6_kernel<4096, true, 4096>;
       else if (gpos && cache == 1024) kfn = tokenize_v6_kernel<1024, true, 4096>;
       else if (gpos && schunk == 128)
@@ -275,7 +284,7 @@ void tokenize_cache_spill(
         cpg = u64p(cpos_g);
       }
       hipLaunchKernelGGL(kfn,
-                         dim3(blocks), dim3(kBlock), 0,
+                         dim3(blocks), dim3(blockn), 0,
                          cur_stream(), text.data_ptr<u8>(), n, (u64)pos_base,
                          u64p(tkeys), tvals.data_ptr<i64>(),
                          texm.numel() ? u64p(texm) : nullptr, (u64)(cap - 1),
