@@ -87,6 +87,25 @@ class WordCountResult:
             out.sort(key=lambda kv: kv[0])
         return out
 
+    def topk(self, k: int) -> List[Tuple[bytes, int]]:
+        """The k most frequent words, descending — device-side torch.topk
+        over counts, then only those k exemplars cross to the host (a
+        serving shortcut the reference would pay a full result read for)."""
+        n = self.counts.numel()
+        k = min(k, n)
+        if k == 0:
+            return []
+        cnt, idx = torch.topk(self.counts, k)
+        lens, blob = ops.extract_words(self.blob_src,
+                                       self.pos.index_select(0, idx))
+        raw = bytes(blob.cpu().numpy().tobytes())
+        out = []
+        off = 0
+        for L, c in zip(lens.cpu().tolist(), cnt.cpu().tolist()):
+            out.append((raw[off:off + L], c))
+            off += L
+        return out
+
 
 class WordCountJob:
     """mode:

@@ -77,6 +77,19 @@ def test_bench_torchrun_ws4_contract():
 
 
 @pytest.mark.timeout(500)
+def test_bench_torchrun_ws8_contract():
+    """8-rank canary — the driver's exact N=8 scaling invocation shape
+    (gloo on CPU here; RCCL on the node shares every line of code)."""
+    d = run_torchrun_bench(8, [
+        "--gpus", "8", "--steps", "1", "--warmup", "1",
+        "--words", "6000", "--splits", "2", "--vocab", "300",
+        "--device", "cpu"])
+    assert d["n_gpus"] == 8
+    assert d["config"]["parallelism"] == "dp8"
+    assert d["config"]["global_batch"] == 48000
+
+
+@pytest.mark.timeout(500)
 def test_bench_torchrun_ws2_contract():
     d = run_torchrun_bench(2, [
         "--gpus", "2", "--steps", "2", "--warmup", "1",

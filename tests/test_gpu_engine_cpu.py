@@ -36,6 +36,13 @@ def test_wordcount_job_single_rank_cpu():
     got = dict(res.to_host())
     exp = counter_oracle(bytes(c.text.numpy().tobytes()))
     assert got == dict(exp)
+    # topk serving shortcut: counts match the oracle's most_common
+    # (tie order is arbitrary — compare count multisets + membership)
+    top = res.topk(10)
+    oracle_top = exp.most_common(10)
+    assert [c for _, c in top] == [c for _, c in oracle_top]
+    for w, c in top:
+        assert exp[w] == c
 
 
 def _dist_worker(rank, world, port, tmpdir):

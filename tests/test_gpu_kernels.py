@@ -253,6 +253,11 @@ def test_streaming_wordcount_mode_vs_counter(dev):
     # lexicographic materialization option
     lex = res.to_host(order="lex")
     assert [w for w, _ in lex] == sorted(got)
+    # topk serving shortcut on device
+    top = res.topk(5)
+    assert [c for _, c in top] == [c for _, c in exp.most_common(5)]
+    for w, c in top:
+        assert exp[w] == c
     # A/B: fused mode must produce identical counts
     job2 = WordCountJob(dev, vocab_estimate=6000, mode="fused")
     res2 = job2.run(text)
