@@ -727,13 +727,14 @@ __global__ __launch_bounds__(BLOCKN) void tokenize_v6_kernel(
     // 32 bytes in registers: my window + 16B lookahead (halo-staged)
     const uint4 va = wactive ? *(const uint4*)&tile[my0] : uint4{0, 0, 0, 0};
     const uint4 vb = wactive ? *(const uint4*)&tile[my0 + 16] : uint4{0, 0, 0, 0};
-    const u32 rs[8] = {va.x, va.y, va.z, va.w, vb.x, vb.y, vb.z, vb.w};
-    u32 m32 = 0;
-    #pragma unroll
-    for (int r = 0; r < 8; ++r)
-      #pragma unroll
-      for (int b = 0; b < 4; ++b)
-        m32 |= (u32)is_ws((u8)(rs[r] >> (8 * b))) << (r * 4 + b);
+    const u64 q0 = (u64)va.x | ((u64)va.y << 32);
+    const u64 q1 = (u64)va.z | ((u64)va.w << 32);
+    const u64 q2 = (u64)vb.x | ((u64)vb.y << 32);
+    const u64 q3 = (u64)vb.z | ((u64)vb.w << 32);
+    // SWAR classify: 4 x ws_mask8 (~17 u64 ops each) replaces 32 per-byte
+    // is_ws calls (ablation: stage+classify was 44% of the kernel at 4 MB)
+    u32 m32 = ws_mask8(q0) | (ws_mask8(q1) << 8) |
+              (ws_mask8(q2) << 16) | (ws_mask8(q3) << 24);
     // bytes at/after `want` count as whitespace (stale stage data)
     long lim32 = want - my0;
     if (lim32 < 32)
@@ -747,10 +748,6 @@ __global__ __launch_bounds__(BLOCKN) void tokenize_v6_kernel(
     if (limw < 16) sm &= (limw <= 0) ? 0u : ((1u << limw) - 1);
     my_words += __popc(sm);
     if (MODE == 1) { my_words += m32; sm = 0; }
-    const u64 q0 = (u64)va.x | ((u64)va.y << 32);
-    const u64 q1 = (u64)va.z | ((u64)va.w << 32);
-    const u64 q2 = (u64)vb.x | ((u64)vb.y << 32);
-    const u64 q3 = (u64)vb.z | ((u64)vb.w << 32);
     // fully-unrolled word loop (a 16-byte window holds <= 8 words):
     // compile-time slot indices keep the miss buffers in REGISTERS —
     // runtime-indexed arrays here lower to divergent v_movrel waterfalls

@@ -264,6 +264,25 @@ def test_streaming_wordcount_mode_vs_counter(dev):
     assert dict(res2.to_host()) == got
 
 
+def test_tokenize_random_binary_all_byte_values(dev):
+    """Whitespace classification over ALL 256 byte values (the SWAR
+    ws_mask8 path must match bytes.split(): ws = {9..13, 32}; bytes
+    >= 128 and control chars are word bytes)."""
+    from mapreduce_amd.gpu.wordcount import WordCountJob
+    rng = np.random.default_rng(77)
+    data = rng.integers(0, 256, size=300_000, dtype=np.uint8)
+    # salt in extra whitespace so words stay shortish
+    data[rng.random(data.size) < 0.12] = 32
+    raw = data.tobytes()
+    text = torch.from_numpy(data.copy()).to(dev)
+    job = WordCountJob(dev, vocab_estimate=1 << 17, mode="streaming")
+    res = job.run(text)
+    exp = collections.Counter(raw.split())
+    assert res.nwords == sum(exp.values())
+    got = dict(res.to_host())
+    assert got == dict(exp)
+
+
 def test_bucketed_vs_chunked_spill_equivalence(dev, monkeypatch):
     """MR_TOK_BSPILL=1 (per-bucket direct spill, no radix bucketize) and
     =0 (wave-chunked spill + radix_pass(56)) must both match the oracle
