@@ -194,10 +194,13 @@ class InvertedIndexJob:
             bucket_off = torch.zeros(257, dtype=torch.int64, device=dev)
             torch.cumsum(totals, 0, out=bucket_off[1:])
             table = ops.make_table(max(1 << 16, k2.numel() // 6), dev)
-            # 2048 LDS slots: composite (word,doc) keys run ~3-5k distinct
-            # per slice — the smaller wordcount table would overflow into
-            # the per-element ht_add fallback
-            ops.ext().bucket_count(hk, pv, bucket_off, 256, 32,
+            # 2048 LDS slots: composite (word,doc) keys are distinct-heavy
+            # (tf ~6.6 -> ~2.4k distinct per 32-slice); 64 slices bring
+            # distinct/slice under the table size so the per-element
+            # ht_add overflow fallback stays cold (A/B: 32/64/128)
+            import os
+            slices = int(os.environ.get("MR_II_SLICES", "64"))
+            ops.ext().bucket_count(hk, pv, bucket_off, 256, slices,
                                    table.tkeys, table.tvals, table.texm,
                                    0, 2048)
             uk2, tf, upos = table.extract()
