@@ -241,7 +241,7 @@ class WordCountJob:
     def _finish_map_inner(self) -> int:
         if self.mode == "streaming":
             import os
-            slices = int(os.environ.get("MR_BUCKET_SLICES", "32"))
+            slices = int(os.environ.get("MR_BUCKET_SLICES", "64"))
             if self._bspill:
                 # misses are already bucket-partitioned in per-bucket
                 # regions; counters hold exact per-bucket lengths
@@ -275,10 +275,10 @@ class WordCountJob:
                 bucket_off = torch.zeros(257, dtype=torch.int64,
                                          device=self.device)
                 torch.cumsum(totals, 0, out=bucket_off[1:])
-                # 32 slices/bucket: the HT_EMPTY chunk padding all lands
-                # in bucket 255 (top byte 0xFF), so fine slicing keeps its
-                # blocks off the critical path (sweep: 8=3.19, 16=2.58,
-                # 32=2.38, 64=2.40 ms/step)
+                # 64 slices/bucket (re-swept at 1024-slot tables:
+                # 16=1.73, 32=1.60, 48=1.55, 64=1.53, 96=1.56 ms/step —
+                # smaller slices raise block-level parallelism and keep
+                # bucket-255's pad skew off the critical path)
                 # 1024 LDS slots: ~390 distinct/slice fits 2.6x over;
                 # 20 KB = 8 blocks/CU (sweep: 2048=1.71, 1024=1.60,
                 # 512=1.64 ms/step — the kernel is occupancy/latency
