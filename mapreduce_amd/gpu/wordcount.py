@@ -2,12 +2,16 @@
 
 Per step (= one full MapReduce job over the rank's corpus):
 
-  MAP+COMBINE   tokenize_count kernel per split: words stream from HBM text
-                straight into the per-rank hash table (the reference's
-                mapfn emit + inline combiner job.lua:83-97, fused into one
-                kernel pass; word count accumulates on-device).
-  EXTRACT+SORT  table -> unique (hash, count, exemplar pos); radix sort by
-                hash (K1) — with mulhi partitioning, the sorted array is
+  MAP+COMBINE   tokenize_v6 kernel over the coalesced splits: per-block
+                LDS cache tables count the Zipf head in-kernel (the
+                reference's mapfn emit + inline combiner job.lua:83-97);
+                cache misses spill through the wave-chunked allocator,
+                then one radix_pass(56) bucketize + per-bucket LDS count
+                (bucket_count) drains them — every per-word atomic is an
+                LDS atomic.
+  EXTRACT+SORT  table -> unique (hash, count, exemplar pos); sort by hash
+                (K1; sub-1M arrays via one stable torch.sort dispatch) —
+                with mulhi partitioning, the sorted array is
                 partition-contiguous.
   SHUFFLE       RCCL all-to-all of (hash, count, exemplar len/bytes) slices
                 (C5/C6): one collective per array over the 7 xGMI links.
