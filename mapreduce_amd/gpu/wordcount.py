@@ -42,6 +42,33 @@ class WordCountResult:
     pos: torch.Tensor       # packed (off<<16|len) into blob_src
     blob_src: torch.Tensor  # u8 source for exemplar bytes
     nwords: int             # words processed by this rank this step
+    hash_kind: str = "wordhash64"  # GPU tier: wordhash64; CPU test tier:
+                                   # fnv1a64 (ops/_cpu.py tokenize_words)
+
+    def key_of(self, word) -> int:
+        """The int64 bit pattern this result keys `word` under (serving
+        lookups must hash with the tier that built the result)."""
+        if isinstance(word, str):
+            word = word.encode()
+        if self.hash_kind == "wordhash64":
+            from mapreduce_amd.utils.tuple import wordhash64
+            k = wordhash64(word)
+        else:
+            k = 0xCBF29CE484222325
+            for b in word:
+                k = ((k ^ b) * 0x100000001B3) & 0xFFFFFFFFFFFFFFFF
+        return k - (1 << 64) if k >= (1 << 63) else k
+
+    def count_of(self, word) -> int:
+        """Point lookup: one binary search on the hash-sorted keys."""
+        ki = self.key_of(word)
+        sk = self.keys ^ (-1 << 63)  # unsigned order -> int64 order
+        q = torch.tensor([ki ^ (-1 << 63)], dtype=torch.int64,
+                         device=sk.device)
+        i = int(torch.searchsorted(sk, q).item())
+        if i >= self.keys.numel() or int(self.keys[i].item()) != ki:
+            return 0
+        return int(self.counts[i].item())
 
     def materialize(self, blocking: bool = True):
         """Deliver the job's results to host memory (the analogue of the
@@ -386,5 +413,6 @@ class WordCountJob:
 
         self._mark("shuffle_reduce")
         self._collect_timing()
-        return WordCountResult(keys=fk, counts=fv, pos=fp,
-                               blob_src=blob_src, nwords=nwords)
+        return WordCountResult(
+            keys=fk, counts=fv, pos=fp, blob_src=blob_src, nwords=nwords,
+            hash_kind="wordhash64" if dev.type == "cuda" else "fnv1a64")

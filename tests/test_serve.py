@@ -1,0 +1,74 @@
+"""HTTP result serving (mapreduce_amd.serve): point queries over finished
+jobs — counts, topk, postings — without materializing results."""
+
+import collections
+
+import numpy as np
+import pytest
+import torch
+
+fastapi = pytest.importorskip("fastapi")
+from fastapi.testclient import TestClient  # noqa: E402
+
+from mapreduce_amd.gpu.corpus import make_corpus
+from mapreduce_amd.gpu.inverted_index import InvertedIndexJob
+from mapreduce_amd.gpu.wordcount import WordCountJob
+from mapreduce_amd.serve import make_app
+
+
+@pytest.fixture(scope="module")
+def served():
+    c = make_corpus("cpu", nwords=30_000, nsplits=4, vocab_size=900, seed=2)
+    wc = WordCountJob("cpu", vocab_estimate=2000).run(c.text, c.splits())
+    rng = np.random.default_rng(5)
+    vocab = [f"w{i}".encode() for i in range(100)]
+    docs = [b" ".join(vocab[i] for i in rng.integers(0, 100, 300).tolist())
+            for _ in range(5)]
+    blob = b" ".join(docs) + b" "
+    offs = [0]
+    for d in docs[:-1]:
+        offs.append(offs[-1] + len(d) + 1)
+    offs.append(len(blob))
+    text = torch.from_numpy(np.frombuffer(blob, dtype=np.uint8).copy())
+    idx = InvertedIndexJob("cpu").run(text, list(zip(offs[:-1], offs[1:])))
+    oracle = collections.Counter(bytes(c.text.numpy().tobytes()).split())
+    return TestClient(make_app(wordcount=wc, index=idx)), oracle, docs
+
+
+def test_healthz(served):
+    client, _, _ = served
+    r = client.get("/healthz").json()
+    assert r == {"ok": True, "wordcount": True, "index": True}
+
+
+def test_count_endpoint(served):
+    client, oracle, _ = served
+    for w in list(oracle)[:20]:
+        r = client.get("/count", params={"word": w.decode()}).json()
+        assert r["count"] == oracle[w], w
+    assert client.get("/count", params={"word": "zzz-absent"}).json() == \
+        {"word": "zzz-absent", "count": 0}
+
+
+def test_topk_endpoint(served):
+    client, oracle, _ = served
+    r = client.get("/topk", params={"k": 5}).json()["topk"]
+    exp = oracle.most_common(5)
+    assert [e["count"] for e in r] == [c for _, c in exp]
+    for e in r:
+        assert oracle[e["word"].encode()] == e["count"]
+
+
+def test_postings_endpoint(served):
+    client, _, docs = served
+    word = b"w7"
+    exp = [(i, doc.split().count(word)) for i, doc in enumerate(docs)
+           if word in doc.split()]
+    r = client.get("/postings", params={"word": "w7"}).json()
+    assert [(p["doc"], p["tf"]) for p in r["postings"]] == exp
+
+
+def test_unmounted_404():
+    client = TestClient(make_app())
+    assert client.get("/count", params={"word": "x"}).status_code == 404
+    assert client.get("/postings", params={"word": "x"}).status_code == 404
