@@ -258,6 +258,10 @@ def test_streaming_wordcount_mode_vs_counter(dev):
     assert [c for _, c in top] == [c for _, c in exp.most_common(5)]
     for w, c in top:
         assert exp[w] == c
+    # point lookups on the device-resident result
+    for w in list(exp)[:5]:
+        assert res.count_of(w) == exp[w]
+    assert res.count_of(b"absent-word-xq") == 0
     # A/B: fused mode must produce identical counts
     job2 = WordCountJob(dev, vocab_estimate=6000, mode="fused")
     res2 = job2.run(text)
