@@ -73,3 +73,47 @@ def serve(wordcount=None, index=None, host: str = "127.0.0.1",
     import uvicorn
 
     uvicorn.run(app or make_app(wordcount, index), host=host, port=port)
+
+
+def build_results_from_files(paths, device=None):
+    """Run wordcount + inverted index over files and return the two
+    result objects (the CLI's one-stop build; GPU if available)."""
+    import torch
+
+    from .gpu.input import load_corpus
+    from .gpu.inverted_index import InvertedIndexJob
+    from .gpu.wordcount import WordCountJob
+
+    dev = device or ("cuda:0" if torch.cuda.is_available() else "cpu")
+    corpus = load_corpus(list(paths), dev)
+    splits = corpus.splits()
+    wc = WordCountJob(dev, vocab_estimate=1 << 17).run(corpus.text, splits)
+    ix = InvertedIndexJob(dev).run(corpus.text, splits)
+    return wc, ix
+
+
+def main(argv: Optional[list] = None) -> int:
+    """CLI: index files and serve queries.
+
+        python -m mapreduce_amd.serve doc1.txt doc2.txt --port 8000
+    """
+    import argparse
+
+    p = argparse.ArgumentParser(description=__doc__)
+    p.add_argument("files", nargs="+")
+    p.add_argument("--host", default="127.0.0.1")
+    p.add_argument("--port", type=int, default=8000)
+    p.add_argument("--device", default=None)
+    args = p.parse_args(argv)
+    wc, ix = build_results_from_files(args.files, args.device)
+    print(f"# serving {wc.nwords or '?'} words, "
+          f"{wc.keys.numel()} unique, {len(args.files)} docs "
+          f"on {args.host}:{args.port}")
+    serve(wc, ix, host=args.host, port=args.port)
+    return 0
+
+
+if __name__ == "__main__":
+    import sys
+
+    sys.exit(main())

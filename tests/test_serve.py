@@ -72,3 +72,19 @@ def test_unmounted_404():
     client = TestClient(make_app())
     assert client.get("/count", params={"word": "x"}).status_code == 404
     assert client.get("/postings", params={"word": "x"}).status_code == 404
+
+
+def test_build_from_files_and_query(tmp_path):
+    """CLI build path: files -> wordcount + index -> queryable app."""
+    from mapreduce_amd.serve import build_results_from_files
+
+    p1 = tmp_path / "a.txt"
+    p2 = tmp_path / "b.txt"
+    p1.write_text("alpha beta alpha gamma\n")
+    p2.write_text("beta beta delta\n")
+    wc, ix = build_results_from_files([str(p1), str(p2)], device="cpu")
+    client = TestClient(make_app(wordcount=wc, index=ix))
+    assert client.get("/count", params={"word": "beta"}).json()["count"] == 3
+    assert client.get("/count", params={"word": "alpha"}).json()["count"] == 2
+    po = client.get("/postings", params={"word": "beta"}).json()["postings"]
+    assert [(p["doc"], p["tf"]) for p in po] == [(0, 1), (1, 2)]
