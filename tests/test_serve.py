@@ -88,3 +88,23 @@ def test_build_from_files_and_query(tmp_path):
     assert client.get("/count", params={"word": "alpha"}).json()["count"] == 2
     po = client.get("/postings", params={"word": "beta"}).json()["postings"]
     assert [(p["doc"], p["tf"]) for p in po] == [(0, 1), (1, 2)]
+
+
+@pytest.mark.gpu
+def test_serve_gpu_results(tmp_path):
+    """Results built on the GPU tier stay device-resident while served."""
+    from mapreduce_amd.serve import build_results_from_files
+
+    p1 = tmp_path / "a.txt"
+    p2 = tmp_path / "b.txt"
+    p1.write_text(("red blue red green " * 500) + "\n")
+    p2.write_text(("blue blue yellow " * 400) + "\n")
+    wc, ix = build_results_from_files([str(p1), str(p2)], device="cuda:0")
+    assert wc.keys.is_cuda and ix.keys.is_cuda
+    client = TestClient(make_app(wordcount=wc, index=ix))
+    assert client.get("/count", params={"word": "red"}).json()["count"] == 1000
+    assert client.get("/count", params={"word": "blue"}).json()["count"] == 1300
+    po = client.get("/postings", params={"word": "blue"}).json()["postings"]
+    assert [(p["doc"], p["tf"]) for p in po] == [(0, 500), (1, 800)]
+    top = client.get("/topk", params={"k": 2}).json()["topk"]
+    assert [t["word"] for t in top] == ["blue", "red"]
