@@ -143,3 +143,17 @@ def test_merge_iterator_many_sorted_runs():
             truth.setdefault(k, []).append(k)
     merged = dict(merge_iterator([iter(r) for r in runs]))
     assert {k: sorted(v) for k, v in merged.items()} == truth
+
+
+def test_module_loader_memoized():
+    """Task-script module loading is memoized per process (the reference's
+    job.lua:387-394 memoizer): repeat loads return the identical object,
+    so init() effects and module state persist across jobs."""
+    from mapreduce_amd.job import load_module as _load_spec
+
+    a = _load_spec("mapreduce_amd.examples.wordcount")
+    b = _load_spec("mapreduce_amd.examples.wordcount")
+    assert a is b
+    assert _load_spec("nil") is None and _load_spec(None) is None
+    d = {"mapfn": lambda k, v, e: None}
+    assert _load_spec(d) is d
