@@ -80,13 +80,20 @@ def exchange(data: torch.Tensor, send_counts: List[int],
                 f"HBM ({free / 1e9:.1f} GB) — partition skew; increase the "
                 "partition count or enable host spill")
     out = torch.empty(need, dtype=data.dtype, device=data.device)
+    if dist.get_backend(group) != "gloo":
+        # RCCL has alltoall — an exception here is a REAL comm failure
+        # and must propagate (a silent P2P retry could hang or corrupt)
+        dist.all_to_all_single(out, data.contiguous(),
+                               output_split_sizes=recv_counts,
+                               input_split_sizes=send_counts, group=group)
+        return out
     try:
         dist.all_to_all_single(out, data.contiguous(),
                                output_split_sizes=recv_counts,
                                input_split_sizes=send_counts, group=group)
         return out
     except (RuntimeError, ValueError):
-        pass
+        pass  # older gloo without alltoall
     # P2P fallback (gloo without alltoall): pairwise rounds
     soff = [0]
     for c in send_counts:
