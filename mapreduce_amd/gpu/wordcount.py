@@ -290,8 +290,10 @@ class WordCountJob:
                         self.table.texm, self._spill_bcap, 1024)
                 self._spill_h = self._spill_p = None
                 return n
-            nspill = int(self._spill_c.item())
-            n = int(self._nwords.item())
+            # ONE packed D2H for both scalars (each .item() is a full
+            # stream sync)
+            both = torch.cat([self._spill_c, self._nwords]).cpu()
+            nspill, n = int(both[0]), int(both[1])
             if nspill > self._spill_cap:
                 # chunk reservations past the cap were dropped in-kernel;
                 # unwritten tail slots would read as garbage keys — fail
