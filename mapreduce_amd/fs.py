@@ -12,7 +12,7 @@ file mediated.  Here:
                      :119-137 (tmpfile + atomic rename like file_builder
                      :80-115)
   * the GPU tier does NOT live here: map outputs stay HBM-resident and move
-    via RCCL all-to-all (mapreduce_amd.gpu.shuffle), replacing C5/C6 of
+    via RCCL all-to-all (mapreduce_amd.gpu.dist.exchange), replacing C5/C6 of
     SURVEY.md §2.5.
 
 Files are streams of length-prefixed pickled (key, values) records sorted by
