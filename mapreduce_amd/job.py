@@ -9,7 +9,7 @@ marks BROKEN on crash (:322-342).
 This module is the GENERAL tier: arbitrary Python user functions, record
 streams on the host.  When a task module declares GPU entry points
 (mapfn_gpu / reduce is a declared-associative+commutative builtin), the GPU
-engine (mapreduce_amd.gpu.engine) takes the whole hot path instead —
+engine (mapreduce_amd.gpu.wordcount + gpu.runner) takes the whole hot path instead —
 mirroring the reference's own fast-path split on reducer property flags
 (job.lua:104-106, 264-274).
 """
@@ -125,7 +125,7 @@ class FnSet:
         self.associative = bool(_get(rmod, "associative_reducer"))
         self.commutative = bool(_get(rmod, "commutative_reducer"))
         self.idempotent = bool(_get(rmod, "idempotent_reducer"))
-        # GPU tier hooks (optional; see mapreduce_amd.gpu.engine)
+        # GPU tier hooks (optional; see mapreduce_amd.gpu.wordcount/runner)
         self.mapfn_gpu = _get(self.modules["mapfn"], "mapfn_gpu")
         self.reducefn_gpu = _get(rmod, "reducefn_gpu")
 
