@@ -44,6 +44,25 @@ def world_info(group=None) -> Tuple[int, int]:
     return 0, 1
 
 
+def exchange_counts_full(send_counts: torch.Tensor,
+                         group=None) -> torch.Tensor:
+    """All-gather of the packed count matrix: send_counts is i64[k*world]
+    (k independent segments back-to-back).  Returns i64[world, k, world]
+    where [i, s, j] = rank i's segment-s count for destination j — every
+    rank sees every rank's send AND receive sizes, so skew detection and
+    chunked-round planning are deterministic across ranks with NO extra
+    collective (the chunked shuffle depends on this: all ranks must agree
+    on the round count or the collectives deadlock)."""
+    rank, world = world_info(group)
+    assert send_counts.numel() % max(world, 1) == 0
+    segs = send_counts.numel() // max(world, 1)
+    if world == 1:
+        return send_counts.clone().view(1, segs, 1)
+    mat = [torch.zeros_like(send_counts) for _ in range(world)]
+    dist.all_gather(mat, send_counts.contiguous(), group=group)
+    return torch.stack(mat).view(world, segs, world)
+
+
 def exchange_counts(send_counts: torch.Tensor, group=None) -> torch.Tensor:
     """Size exchange for the all-to-all (C5): send_counts is i64[k*world]
     (k independent segments packed back-to-back; callers pack several
@@ -52,11 +71,7 @@ def exchange_counts(send_counts: torch.Tensor, group=None) -> torch.Tensor:
     rank, world = world_info(group)
     if world == 1:
         return send_counts.clone()
-    assert send_counts.numel() % world == 0
-    segs = send_counts.numel() // world
-    mat = [torch.zeros_like(send_counts) for _ in range(world)]
-    dist.all_gather(mat, send_counts.contiguous(), group=group)
-    stacked = torch.stack(mat).view(world, segs, world)
+    stacked = exchange_counts_full(send_counts, group)
     return stacked[:, :, rank].transpose(0, 1).reshape(-1)
 
 

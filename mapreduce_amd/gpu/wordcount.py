@@ -164,6 +164,8 @@ class WordCountJob:
         self._hip_events = timing and self.device.type == "cuda"
         self.last_phase_ms: dict = {}
         self._events: list = []
+        # rounds the last shuffle used (>1 = skew guard chunked it)
+        self.last_shuffle_rounds = 1
 
     def _mark(self, name: str) -> None:
         if not self.timing:
@@ -356,11 +358,128 @@ class WordCountJob:
         nwords = self.finish_map()
         return self.shuffle_reduce(nwords)
 
+    def _shuffle_budget_bytes(self) -> int:
+        """Per-rank receive budget for one shuffle round.  MUST be
+        deterministic and identical across ranks (all ranks derive the
+        round count from it — disagreement would deadlock the
+        collectives), so the auto value uses TOTAL device memory, never
+        free memory.  `MR_SHUFFLE_BUDGET_BYTES` overrides (0 disables
+        chunking); default on GPU is total HBM / 4 (72 GB on MI355X —
+        receive + sort scratch + accumulator fit with headroom), and on
+        the CPU test tier chunking runs only when the env forces it."""
+        import os
+        env = os.environ.get("MR_SHUFFLE_BUDGET_BYTES")
+        if env is not None:
+            return max(0, int(env))
+        if self.device.type == "cuda":
+            props = torch.cuda.get_device_properties(self.device)
+            return int(props.total_memory) // 4
+        return 0
+
+    def _plan_shuffle_rounds(self, full: torch.Tensor) -> int:
+        """Round count for the chunked shuffle (1 = single-shot).
+        `full` is the host copy of exchange_counts_full's [i, seg, j]
+        matrix (seg 0 = element counts, seg 1 = blob bytes), identical
+        on every rank, so every rank computes the same answer.  The
+        bound is the LARGEST rank's receive (the skewed partition's
+        owner defines the round count for everyone — SURVEY.md §7
+        "shuffle skew + memory budget")."""
+        budget = self._shuffle_budget_bytes()
+        if budget <= 0:
+            return 1
+        # 3 interleaved i64 arrays (key, count, len) + exemplar bytes
+        need = (full[:, 0, :].sum(0) * 24 + full[:, 1, :].sum(0))
+        worst = int(need.max().item())
+        if worst <= budget:
+            return 1
+        # cap: even a pathological budget terminates; each round still
+        # moves >= 1 element per (src, dst) pair
+        return min((worst + budget - 1) // budget, 4096)
+
+    def _chunked_shuffle_reduce(self, sk, sv, lens, blob, lcs,
+                                full: torch.Tensor, rounds: int):
+        """Bounded-memory shuffle: the all-to-all runs in `rounds`
+        passes, each moving a 1/rounds slice of every partition segment,
+        and each received slice is folded into a running reduced
+        accumulator (sort + segmented reduce + blob compaction).  Exact
+        for the declared associative+commutative reducer — the same
+        property the reference's combiner fast path relies on
+        (job.lua:264-274) — because count merging is order-free.  HBM
+        high-water mark: one round's receive + the unique-key
+        accumulator, instead of the full skewed partition (SURVEY.md §7;
+        the reference never hits this because GridFS is unbounded)."""
+        dev = self.device
+        world, rank = self.world, self.rank
+        send_c = full[rank, 0].tolist()
+        # element boundaries: partition p, round r starts at
+        # poff[p] + c_p * r // rounds  (host ints — no device sync)
+        poff = [0]
+        for c in send_c:
+            poff.append(poff[-1] + c)
+        ebnd = [[poff[p] + send_c[p] * r // rounds
+                 for r in range(rounds + 1)] for p in range(world)]
+        # byte offset of each element boundary: ecs[e] (start of element
+        # e's exemplar bytes), with e == n mapping to total blob bytes
+        total_b = int(blob.numel())
+        ecs = torch.cat([lcs - lens,
+                         torch.tensor([total_b], dtype=torch.int64,
+                                      device=dev)])
+        bidx = torch.tensor([e for row in ebnd for e in row],
+                            dtype=torch.int64, device=dev)
+        boff = ecs.index_select(0, bidx).cpu().view(world, rounds + 1)
+        bbnd = boff.tolist()
+        # per-round blob send sizes -> ONE packed counts exchange
+        # (element counts per round need no exchange: every rank derives
+        # them from `full`'s per-partition counts with the same formula)
+        rb = torch.tensor(
+            [bbnd[p][r + 1] - bbnd[p][r]
+             for r in range(rounds) for p in range(world)],
+            dtype=torch.int64, device=dev)
+        recv_rb = (dx.exchange_counts(rb, self.group)
+                   .cpu().view(rounds, world).tolist())
+        cin = full[:, 0, rank].tolist()  # elements each source sends me
+        tri2d = torch.stack([sk, sv, lens], dim=1)
+
+        acc_k = torch.empty(0, dtype=torch.int64, device=dev)
+        acc_v = torch.empty(0, dtype=torch.int64, device=dev)
+        acc_p = torch.empty(0, dtype=torch.int64, device=dev)
+        acc_blob = torch.empty(0, dtype=torch.uint8, device=dev)
+        for r in range(rounds):
+            sc = [ebnd[p][r + 1] - ebnd[p][r] for p in range(world)]
+            rc = [cin[i] * (r + 1) // rounds - cin[i] * r // rounds
+                  for i in range(world)]
+            sb = [bbnd[p][r + 1] - bbnd[p][r] for p in range(world)]
+            stri = torch.cat([tri2d[ebnd[p][r]:ebnd[p][r + 1]]
+                              for p in range(world)]).reshape(-1)
+            sblob = torch.cat([blob[bbnd[p][r]:bbnd[p][r + 1]]
+                               for p in range(world)])
+            rtri = dx.exchange(stri, [3 * c for c in sc],
+                               [3 * c for c in rc], self.group).view(-1, 3)
+            rblob = dx.exchange(sblob, sb, recv_rb[r], self.group)
+            rlens = rtri[:, 2].contiguous()
+            roff = torch.cumsum(rlens, 0) - rlens
+            rpos = ((roff + int(acc_blob.numel())) << 16) | rlens
+            # fold into the accumulator: counts add, first exemplar kept
+            mk = torch.cat([acc_k, rtri[:, 0].contiguous()])
+            mv = torch.cat([acc_v, rtri[:, 1].contiguous()])
+            mp = torch.cat([acc_p, rpos])
+            acc_blob = torch.cat([acc_blob, rblob])
+            k2, v2, p2 = ops.sort_by_key(mk, mv, mp)
+            acc_k, acc_v, acc_p, _ = ops.reduce_by_key_sorted(k2, v2, p2)
+            # compact: keep only surviving exemplars' bytes so the blob
+            # stays O(unique keys), not O(rounds x receive)
+            clens, acc_blob = ops.extract_words(acc_blob, acc_p)
+            clens = clens.to(dev)
+            coff = torch.cumsum(clens, 0) - clens
+            acc_p = (coff << 16) | clens
+        return acc_k, acc_v, acc_p, acc_blob
+
     def shuffle_reduce(self, nwords: int) -> WordCountResult:
         """Phase 2: extract + sort uniques, all-to-all exchange, segmented
         reduce — collective across ranks (every rank must enter)."""
         text = self._text
         dev = self.device
+        self.last_shuffle_rounds = 1
 
         # ---- EXTRACT + SORT
         uk, uv, up = self.table.extract()
@@ -379,36 +498,45 @@ class WordCountJob:
                     torch.zeros_like(bnd))
                 blob_counts_d = torch.cat([cum[:1], cum[1:] - cum[:-1]])
             else:
+                lcs = torch.zeros(0, dtype=torch.int64, device=dev)
                 blob_counts_d = torch.zeros_like(counts_d)
             # ONE packed size exchange + ONE host sync for both arrays
             # (xGMI collectives and D2H syncs are per-call latency-bound;
-            # fewer+larger wins — guide)
+            # fewer+larger wins — guide).  The FULL count matrix comes
+            # back, so every rank can compute every rank's receive size
+            # and agree on the chunked-round count with no extra
+            # collective.
             packed = torch.cat([counts_d, blob_counts_d])
-            recv_packed = dx.exchange_counts(packed, self.group)
-            host = torch.stack([packed, recv_packed]).cpu()
-            send_c = host[0, :self.world].tolist()
-            send_b = host[0, self.world:].tolist()
-            recv_c = host[1, :self.world].tolist()
-            recv_b = host[1, self.world:].tolist()
-            # ONE i64 all-to-all for (key, count, len): rows interleave the
-            # three arrays, so each partition's segment stays contiguous
-            # and split sizes just triple
-            tri = torch.stack([sk, sv, lens], dim=1).reshape(-1)
-            rtri = dx.exchange(tri, [3 * c for c in send_c],
-                               [3 * c for c in recv_c], self.group)
-            rtri = rtri.view(-1, 3)
-            rk = rtri[:, 0].contiguous()
-            rv = rtri[:, 1].contiguous()
-            rlens = rtri[:, 2].contiguous()
-            rblob = dx.exchange(blob, send_b, recv_b, self.group)
-            # rebuild packed positions into the received blob
-            roff = torch.cumsum(rlens, 0) - rlens
-            rpos = (roff << 16) | rlens
+            full = dx.exchange_counts_full(packed, self.group).cpu()
+            send_c = full[self.rank, 0].tolist()
+            send_b = full[self.rank, 1].tolist()
+            recv_c = full[:, 0, self.rank].tolist()
+            recv_b = full[:, 1, self.rank].tolist()
+            rounds = self._plan_shuffle_rounds(full)
+            self.last_shuffle_rounds = rounds
+            if rounds > 1:
+                fk, fv, fp, blob_src = self._chunked_shuffle_reduce(
+                    sk, sv, lens, blob, lcs, full, rounds)
+            else:
+                # ONE i64 all-to-all for (key, count, len): rows interleave
+                # the three arrays, so each partition's segment stays
+                # contiguous and split sizes just triple
+                tri = torch.stack([sk, sv, lens], dim=1).reshape(-1)
+                rtri = dx.exchange(tri, [3 * c for c in send_c],
+                                   [3 * c for c in recv_c], self.group)
+                rtri = rtri.view(-1, 3)
+                rk = rtri[:, 0].contiguous()
+                rv = rtri[:, 1].contiguous()
+                rlens = rtri[:, 2].contiguous()
+                rblob = dx.exchange(blob, send_b, recv_b, self.group)
+                # rebuild packed positions into the received blob
+                roff = torch.cumsum(rlens, 0) - rlens
+                rpos = (roff << 16) | rlens
 
-            # ---- REDUCE: sort received runs, segment, first-exemplar
-            k2, v2, p2 = ops.sort_by_key(rk, rv, rpos)
-            fk, fv, fp, _ = ops.reduce_by_key_sorted(k2, v2, p2)
-            blob_src = rblob
+                # ---- REDUCE: sort received runs, segment, first-exemplar
+                k2, v2, p2 = ops.sort_by_key(rk, rv, rpos)
+                fk, fv, fp, _ = ops.reduce_by_key_sorted(k2, v2, p2)
+                blob_src = rblob
         else:
             fk, fv, fp = sk, sv, sp
             blob_src = text

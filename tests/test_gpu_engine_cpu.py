@@ -90,6 +90,54 @@ def test_wordcount_job_gloo_ws2(tmp_path):
         _dist_worker, args=(2, port, str(tmp_path)), nprocs=2, join=True)
 
 
+def _chunked_worker(rank, world, port):
+    """Skew/memory guard: a tiny MR_SHUFFLE_BUDGET_BYTES forces the
+    shuffle into multiple bounded-memory rounds; results must be
+    byte-identical to the single-shot exchange (the round-folded
+    reduce is exact for the assoc+comm sum reducer)."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.distributed.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        c = make_corpus("cpu", nwords=6_000, nsplits=3, vocab_size=500,
+                        seed=7 + rank)
+        # single-shot reference first (budget disabled)
+        os.environ["MR_SHUFFLE_BUDGET_BYTES"] = "0"
+        job = WordCountJob("cpu", vocab_estimate=1000)
+        ref = job.run(c.text, c.splits())
+        assert job.last_shuffle_rounds == 1
+        ref_pairs = sorted(ref.to_host())
+        # now force many rounds; far below the ~recv size so rounds > 1
+        os.environ["MR_SHUFFLE_BUDGET_BYTES"] = "4096"
+        job2 = WordCountJob("cpu", vocab_estimate=1000)
+        got = job2.run(c.text, c.splits())
+        assert job2.last_shuffle_rounds > 1, job2.last_shuffle_rounds
+        # every rank agreed on the round count (a disagreement would
+        # have deadlocked the collectives above, but assert anyway)
+        rt = torch.tensor([job2.last_shuffle_rounds])
+        mx = rt.clone()
+        torch.distributed.all_reduce(mx, op=torch.distributed.ReduceOp.MAX)
+        assert int(mx.item()) == job2.last_shuffle_rounds
+        assert got.nwords == ref.nwords
+        assert sorted(got.to_host()) == ref_pairs
+        # count_of serving lookup still works on the chunked result
+        w, n = ref_pairs[0]
+        assert got.count_of(w) == n
+    finally:
+        os.environ.pop("MR_SHUFFLE_BUDGET_BYTES", None)
+        torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_wordcount_chunked_shuffle_gloo_ws2():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    torch.multiprocessing.spawn(_chunked_worker, args=(2, port), nprocs=2,
+                                join=True)
+
+
 def _timing_worker(rank, world, port):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
