@@ -193,6 +193,10 @@ class GpuClusterRunner:
         phase_ms = getattr(self.job, "last_phase_ms", None)
         if phase_ms:
             extra["phase_ms"] = {k: round(v, 4) for k, v in phase_ms.items()}
+        rounds = getattr(self.job, "last_shuffle_rounds", None)
+        if rounds is not None:
+            # >1 = the skew/memory guard chunked the last shuffle
+            extra["shuffle_rounds"] = rounds
         if self.claim_mode == "batch":
             doc, _ = self.coord.get_doc(f"{ns}/batch")
             return {"jobs": doc["nsplits"] if doc else 0,
