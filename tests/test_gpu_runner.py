@@ -27,7 +27,8 @@ def test_runner_single_rank(claim_mode):
     assert dict(res.to_host()) == oracle(bytes(c.text.numpy().tobytes()))
     st = runner.job_stats()
     if claim_mode == "dynamic":
-        assert st == {"jobs": 5, "written": 5, "broken": 0}
+        assert st == {"jobs": 5, "written": 5, "broken": 0,
+                      "shuffle_rounds": 1}
     else:
         assert st["status"] == STATUS.WRITTEN
 
