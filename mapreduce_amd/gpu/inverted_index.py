@@ -56,6 +56,16 @@ class InvertedIndexResult:
                               tf[offs[i]:offs[i + 1]]))
         return out
 
+    def pair_iterator(self, order: str = "hash"):
+        """The reference finalfn contract (server.lua:360-385): yields
+        (word, postings-list) pairs; order="lex" sorts by word bytes at
+        the finalfn boundary (the reference's sorted-result guarantee)."""
+        items = self.to_host().items()
+        if order == "lex":
+            items = sorted(items)
+        for w, postings in items:
+            yield w, postings
+
     def _signed_keys(self) -> torch.Tensor:
         """keys hold u64 bit patterns; XOR the sign bit maps unsigned
         order onto int64 order so torch.searchsorted works (cached)."""

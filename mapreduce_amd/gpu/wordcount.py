@@ -118,6 +118,14 @@ class WordCountResult:
             out.sort(key=lambda kv: kv[0])
         return out
 
+    def pair_iterator(self, order: str = "hash"):
+        """The reference finalfn contract (server.lua:360-385): yields
+        (key, values) with values a list — here the reduced [count] —
+        so a host-tier finalfn consumes a GPU-tier result unchanged.
+        order as in to_host ("lex" = the reference's sorted guarantee)."""
+        for w, c in self.to_host(order=order):
+            yield w, [c]
+
     def topk(self, k: int) -> List[Tuple[bytes, int]]:
         """The k most frequent words, descending — device-side torch.topk
         over counts, then only those k exemplars cross to the host (a

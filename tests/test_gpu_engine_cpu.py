@@ -176,3 +176,43 @@ def test_cluster_timing_stats_gloo_ws2():
     s.close()
     torch.multiprocessing.spawn(_timing_worker, args=(2, port), nprocs=2,
                                 join=True)
+
+
+def test_pair_iterator_finalfn_contract():
+    """GPU-tier results feed a host-tier finalfn unchanged: pair_iterator
+    yields (key, values-list) pairs (server.lua:360-385 contract)."""
+    c = make_corpus("cpu", nwords=3_000, nsplits=2, vocab_size=200, seed=9)
+    job = WordCountJob("cpu", vocab_estimate=400)
+    res = job.run(c.text, c.splits())
+    exp = counter_oracle(bytes(c.text.numpy().tobytes()))
+    got = {}
+    last = None
+    for k, vs in res.pair_iterator(order="lex"):
+        assert isinstance(vs, list) and len(vs) == 1
+        assert last is None or k > last  # lexicographic guarantee
+        last = k
+        got[k] = vs[0]
+    assert got == dict(exp)
+
+    # a reference-shaped finalfn consuming the iterator
+    seen = {}
+
+    def finalfn(it):
+        for k, vs in it:
+            seen[k] = sum(vs)
+        return True
+
+    assert finalfn(res.pair_iterator()) is True
+    assert seen == dict(exp)
+
+
+def test_inverted_index_pair_iterator():
+    from mapreduce_amd.gpu.inverted_index import InvertedIndexJob
+
+    c = make_corpus("cpu", nwords=2_000, nsplits=4, vocab_size=150, seed=11)
+    job = InvertedIndexJob("cpu")
+    res = job.run(c.text, c.splits())
+    exp = res.to_host()
+    got = dict(res.pair_iterator(order="lex"))
+    assert got == exp
+    assert list(dict(res.pair_iterator(order="lex"))) == sorted(exp)
