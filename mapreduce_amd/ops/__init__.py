@@ -189,13 +189,21 @@ def sort_by_key(keys: torch.Tensor, *others: torch.Tensor, bits: int = 64):
 
 def reduce_by_key_sorted(keys: torch.Tensor,
                          vals: Optional[torch.Tensor] = None,
-                         aux: Optional[torch.Tensor] = None):
+                         aux: Optional[torch.Tensor] = None,
+                         op: str = "sum"):
     """keys sorted (u64 bit order); vals i64/f64 or None (=count 1s);
     aux: optional per-element u64 whose first value per segment is kept
-    (exemplar positions).  Returns (ukeys, reduced, uaux?, nseg)."""
+    (exemplar positions).  op: "sum" (i64/f64), or "min"/"max" (i64) —
+    the canonical associative+commutative(+idempotent) reducers the
+    fast-path property flags admit (job.lua:104-106).
+    Returns (ukeys, reduced, uaux?, nseg)."""
+    if op not in ("sum", "min", "max"):
+        raise ValueError(f"unsupported op {op!r}")
+    if op != "sum" and (vals is None or vals.dtype != torch.int64):
+        raise TypeError("min/max reduction needs i64 vals")
     if not keys.is_cuda:
         from . import _cpu
-        return _cpu.reduce_by_key_sorted(keys, vals, aux)
+        return _cpu.reduce_by_key_sorted(keys, vals, aux, op)
     n = keys.numel()
     if n == 0:
         z = torch.empty(0, dtype=torch.int64, device=keys.device)
@@ -203,7 +211,10 @@ def reduce_by_key_sorted(keys: torch.Tensor,
     flags = ext().head_flags(keys)
     seg = torch.cumsum(flags, 0)
     nseg = int(seg[-1].item())
-    if vals is None or vals.dtype == torch.int64:
+    if op != "sum":
+        uk, uv = ext().seg_reduce_i64_minmax(keys, vals, seg, nseg,
+                                             op == "min")
+    elif vals is None or vals.dtype == torch.int64:
         v = vals if vals is not None else torch.empty(
             0, dtype=torch.int64, device=keys.device)
         uk, uv = ext().seg_reduce_i64(keys, v, seg, nseg)

@@ -99,10 +99,14 @@ def sort_pairs(keys: torch.Tensor, vals: Optional[torch.Tensor], bits: int):
     return sk, vals[torch.from_numpy(order.astype(np.int64))]
 
 
-def reduce_by_key_sorted(keys, vals, aux):
+def reduce_by_key_sorted(keys, vals, aux, op="sum"):
     ku = _u64(keys)
     ukeys, idx = np.unique(ku, return_index=True)
-    if vals is None:
+    if op == "min":
+        uv = torch.from_numpy(np.minimum.reduceat(vals.numpy(), idx))
+    elif op == "max":
+        uv = torch.from_numpy(np.maximum.reduceat(vals.numpy(), idx))
+    elif vals is None:
         sums = np.add.reduceat(np.ones(len(ku), dtype=np.int64), idx)
         uv = torch.from_numpy(sums)
     else:

@@ -1343,6 +1343,29 @@ __global__ void seg_scatter_i64_kernel(const u64* __restrict__ keys,
   }
 }
 
+// min/max variants of the segmented scatter-reduce — the other canonical
+// associative+commutative+idempotent reducers the fast-path property
+// flags admit (job.lua:104-106; min/max are idempotent, so they are safe
+// even under the combiner's re-application).  Signed i64 comparison;
+// ovals must be pre-initialized to INT64_MAX (min) / INT64_MIN (max) by
+// the host wrapper.
+template <bool IS_MIN>
+__global__ void seg_scatter_i64_minmax_kernel(
+    const u64* __restrict__ keys, const i64* __restrict__ vals,
+    const i64* __restrict__ seg, long n, u64* __restrict__ okeys,
+    i64* __restrict__ ovals) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    i64 s = seg[i] - 1;
+    if (i == 0 || seg[i] != seg[i - 1]) okeys[s] = keys[i];
+    if (IS_MIN)
+      atomicMin((long long*)&ovals[s], (long long)vals[i]);
+    else
+      atomicMax((long long*)&ovals[s], (long long)vals[i]);
+  }
+}
+
 // pick one auxiliary value (e.g. exemplar pos) per segment: first element
 __global__ void seg_first_u64_kernel(const u64* __restrict__ aux,
                                      const i64* __restrict__ seg, long n,

@@ -9,6 +9,8 @@
 #include <ATen/hip/HIPContext.h>
 #include <torch/extension.h>
 
+#include <limits>
+
 #include "gradsum.hip"
 #include "mr_kernels.hip"
 #include "radix_sort.hip"
@@ -572,6 +574,34 @@ std::vector<torch::Tensor> seg_reduce_i64(torch::Tensor keys,
   return {okeys, ovals};
 }
 
+std::vector<torch::Tensor> seg_reduce_i64_minmax(torch::Tensor keys,
+                                                 torch::Tensor vals,
+                                                 torch::Tensor seg, long nseg,
+                                                 bool is_min) {
+  check_dev_i64(keys, "keys");
+  long n = keys.numel();
+  auto okeys = torch::empty({nseg}, keys.options());
+  auto ovals = torch::full({nseg},
+                           is_min ? std::numeric_limits<i64>::max()
+                                  : std::numeric_limits<i64>::min(),
+                           keys.options());
+  if (n) {
+    if (is_min)
+      hipLaunchKernelGGL(seg_scatter_i64_minmax_kernel<true>,
+                         dim3(grid_for(n)), dim3(kBlock), 0, cur_stream(),
+                         u64cp(keys), vals.data_ptr<i64>(),
+                         seg.data_ptr<i64>(), n, u64p(okeys),
+                         ovals.data_ptr<i64>());
+    else
+      hipLaunchKernelGGL(seg_scatter_i64_minmax_kernel<false>,
+                         dim3(grid_for(n)), dim3(kBlock), 0, cur_stream(),
+                         u64cp(keys), vals.data_ptr<i64>(),
+                         seg.data_ptr<i64>(), n, u64p(okeys),
+                         ovals.data_ptr<i64>());
+  }
+  return {okeys, ovals};
+}
+
 torch::Tensor seg_first_u64(torch::Tensor aux, torch::Tensor seg, long nseg) {
   long n = aux.numel();
   auto oaux = torch::zeros({nseg}, aux.options());
@@ -736,6 +766,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "chunked compaction (HT_EMPTY-padded; mask/trim downstream)");
   m.def("head_flags", &head_flags);
   m.def("seg_reduce_i64", &seg_reduce_i64);
+  m.def("seg_reduce_i64_minmax", &seg_reduce_i64_minmax);
   m.def("seg_first_u64", &seg_first_u64);
   m.def("seg_reduce_f64", &seg_reduce_f64);
   m.def("partition_hist", &partition_hist);

@@ -399,3 +399,21 @@ def test_gpu_wordcount_pipeline_vs_counter(dev):
     # sortedness of the final key order (u64 bit order)
     ks = u64view(sk.cpu())
     assert np.array_equal(ks, np.sort(ks))
+
+
+def test_reduce_by_key_min_max(dev):
+    """i64 min/max segmented reduce vs NumPy reduceat (negative values
+    included — signed comparison, not the u64 bit order keys use)."""
+    from mapreduce_amd import ops
+    rng = np.random.default_rng(13)
+    n = 300_000
+    keys_np = np.sort(rng.integers(0, 5_000, size=n, dtype=np.uint64) * 7919)
+    vals_np = rng.integers(-10 ** 12, 10 ** 12, size=n, dtype=np.int64)
+    keys = torch.from_numpy(keys_np.view(np.int64)).to(dev)
+    vals = torch.from_numpy(vals_np).to(dev)
+    exp_keys, idx = np.unique(keys_np, return_index=True)
+    for op, red in (("min", np.minimum), ("max", np.maximum)):
+        uk, uv, _, nseg = ops.reduce_by_key_sorted(keys, vals, op=op)
+        assert nseg == len(exp_keys)
+        assert np.array_equal(u64view(uk.cpu()), exp_keys)
+        assert np.array_equal(uv.cpu().numpy(), red.reduceat(vals_np, idx))

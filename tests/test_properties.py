@@ -107,3 +107,28 @@ def test_cpu_sort_by_key_unsigned_order(vals):
     # payload permuted consistently
     orig = np.array(vals, dtype=np.uint64)
     assert (orig[sv.numpy()] == out).all()
+
+
+def test_reduce_by_key_min_max_cpu():
+    """CPU tier of the op= extension (the GPU kernel is diffed against
+    this same reduceat oracle in tests/test_gpu_kernels.py)."""
+    import numpy as np
+    import torch
+
+    from mapreduce_amd import ops
+
+    rng = np.random.default_rng(3)
+    keys_np = np.sort(rng.integers(0, 50, size=2_000, dtype=np.uint64))
+    vals_np = rng.integers(-10 ** 9, 10 ** 9, size=2_000, dtype=np.int64)
+    keys = torch.from_numpy(keys_np.view(np.int64))
+    vals = torch.from_numpy(vals_np)
+    exp_keys, idx = np.unique(keys_np, return_index=True)
+    for op, red in (("min", np.minimum), ("max", np.maximum)):
+        uk, uv, _, nseg = ops.reduce_by_key_sorted(keys, vals, op=op)
+        assert nseg == len(exp_keys)
+        assert np.array_equal(uv.numpy(), red.reduceat(vals_np, idx))
+    import pytest
+    with pytest.raises(TypeError):
+        ops.reduce_by_key_sorted(keys, None, op="min")
+    with pytest.raises(ValueError):
+        ops.reduce_by_key_sorted(keys, vals, op="mean")
