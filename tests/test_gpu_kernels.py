@@ -338,7 +338,8 @@ def test_cluster_runner_dynamic_gpu(dev):
     exp = collections.Counter(bytes(c.text.cpu().numpy().tobytes()).split())
     assert got == dict(exp)
     st = runner.job_stats()
-    assert st == {"jobs": 6, "written": 6, "broken": 0}
+    assert st == {"jobs": 6, "written": 6, "broken": 0,
+                  "shuffle_rounds": 1}
 
 
 def test_streamed_file_wordcount_gpu(dev, tmp_path):
