@@ -193,14 +193,15 @@ def reduce_by_key_sorted(keys: torch.Tensor,
                          op: str = "sum"):
     """keys sorted (u64 bit order); vals i64/f64 or None (=count 1s);
     aux: optional per-element u64 whose first value per segment is kept
-    (exemplar positions).  op: "sum" (i64/f64), or "min"/"max" (i64) —
-    the canonical associative+commutative(+idempotent) reducers the
+    (exemplar positions).  op: "sum"/"min"/"max" (i64 or f64) — the
+    canonical associative+commutative(+idempotent) reducers the
     fast-path property flags admit (job.lua:104-106).
     Returns (ukeys, reduced, uaux?, nseg)."""
     if op not in ("sum", "min", "max"):
         raise ValueError(f"unsupported op {op!r}")
-    if op != "sum" and (vals is None or vals.dtype != torch.int64):
-        raise TypeError("min/max reduction needs i64 vals")
+    if op != "sum" and (vals is None or
+                        vals.dtype not in (torch.int64, torch.float64)):
+        raise TypeError("min/max reduction needs i64 or f64 vals")
     if not keys.is_cuda:
         from . import _cpu
         return _cpu.reduce_by_key_sorted(keys, vals, aux, op)
@@ -212,8 +213,9 @@ def reduce_by_key_sorted(keys: torch.Tensor,
     seg = torch.cumsum(flags, 0)
     nseg = int(seg[-1].item())
     if op != "sum":
-        uk, uv = ext().seg_reduce_i64_minmax(keys, vals, seg, nseg,
-                                             op == "min")
+        fn = (ext().seg_reduce_i64_minmax if vals.dtype == torch.int64
+              else ext().seg_reduce_f64_minmax)
+        uk, uv = fn(keys, vals, seg, nseg, op == "min")
     elif vals is None or vals.dtype == torch.int64:
         v = vals if vals is not None else torch.empty(
             0, dtype=torch.int64, device=keys.device)

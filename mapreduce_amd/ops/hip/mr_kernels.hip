@@ -1366,6 +1366,25 @@ __global__ void seg_scatter_i64_minmax_kernel(
   }
 }
 
+// f64 twin (HIP provides native double atomicMin/Max); ovals
+// pre-initialized to +inf / -inf by the host wrapper
+template <bool IS_MIN>
+__global__ void seg_scatter_f64_minmax_kernel(
+    const u64* __restrict__ keys, const double* __restrict__ vals,
+    const i64* __restrict__ seg, long n, u64* __restrict__ okeys,
+    double* __restrict__ ovals) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    i64 s = seg[i] - 1;
+    if (i == 0 || seg[i] != seg[i - 1]) okeys[s] = keys[i];
+    if (IS_MIN)
+      atomicMin(&ovals[s], vals[i]);
+    else
+      atomicMax(&ovals[s], vals[i]);
+  }
+}
+
 // pick one auxiliary value (e.g. exemplar pos) per segment: first element
 __global__ void seg_first_u64_kernel(const u64* __restrict__ aux,
                                      const i64* __restrict__ seg, long n,

@@ -603,6 +603,34 @@ std::vector<torch::Tensor> seg_reduce_i64_minmax(torch::Tensor keys,
   return {okeys, ovals};
 }
 
+std::vector<torch::Tensor> seg_reduce_f64_minmax(torch::Tensor keys,
+                                                 torch::Tensor vals,
+                                                 torch::Tensor seg, long nseg,
+                                                 bool is_min) {
+  check_dev_i64(keys, "keys");
+  long n = keys.numel();
+  auto okeys = torch::empty({nseg}, keys.options());
+  auto ovals = torch::full({nseg},
+                           is_min ? std::numeric_limits<double>::infinity()
+                                  : -std::numeric_limits<double>::infinity(),
+                           vals.options());
+  if (n) {
+    if (is_min)
+      hipLaunchKernelGGL(seg_scatter_f64_minmax_kernel<true>,
+                         dim3(grid_for(n)), dim3(kBlock), 0, cur_stream(),
+                         u64cp(keys), vals.data_ptr<double>(),
+                         seg.data_ptr<i64>(), n, u64p(okeys),
+                         ovals.data_ptr<double>());
+    else
+      hipLaunchKernelGGL(seg_scatter_f64_minmax_kernel<false>,
+                         dim3(grid_for(n)), dim3(kBlock), 0, cur_stream(),
+                         u64cp(keys), vals.data_ptr<double>(),
+                         seg.data_ptr<i64>(), n, u64p(okeys),
+                         ovals.data_ptr<double>());
+  }
+  return {okeys, ovals};
+}
+
 torch::Tensor seg_first_u64(torch::Tensor aux, torch::Tensor seg, long nseg) {
   long n = aux.numel();
   auto oaux = torch::zeros({nseg}, aux.options());
@@ -768,6 +796,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("head_flags", &head_flags);
   m.def("seg_reduce_i64", &seg_reduce_i64);
   m.def("seg_reduce_i64_minmax", &seg_reduce_i64_minmax);
+  m.def("seg_reduce_f64_minmax", &seg_reduce_f64_minmax);
   m.def("seg_first_u64", &seg_first_u64);
   m.def("seg_reduce_f64", &seg_reduce_f64);
   m.def("partition_hist", &partition_hist);

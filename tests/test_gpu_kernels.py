@@ -418,3 +418,10 @@ def test_reduce_by_key_min_max(dev):
         assert nseg == len(exp_keys)
         assert np.array_equal(u64view(uk.cpu()), exp_keys)
         assert np.array_equal(uv.cpu().numpy(), red.reduceat(vals_np, idx))
+    # f64 twin (native double atomicMin/Max)
+    fvals_np = rng.standard_normal(n)
+    fvals = torch.from_numpy(fvals_np).to(dev)
+    for op, red in (("min", np.minimum), ("max", np.maximum)):
+        uk, uv, _, nseg = ops.reduce_by_key_sorted(keys, fvals, op=op)
+        assert nseg == len(exp_keys)
+        assert np.array_equal(uv.cpu().numpy(), red.reduceat(fvals_np, idx))

@@ -127,6 +127,10 @@ def test_reduce_by_key_min_max_cpu():
         uk, uv, _, nseg = ops.reduce_by_key_sorted(keys, vals, op=op)
         assert nseg == len(exp_keys)
         assert np.array_equal(uv.numpy(), red.reduceat(vals_np, idx))
+        fv = torch.from_numpy(vals_np.astype(np.float64))
+        _, uvf, _, _ = ops.reduce_by_key_sorted(keys, fv, op=op)
+        assert np.array_equal(uvf.numpy(),
+                              red.reduceat(vals_np.astype(np.float64), idx))
     import pytest
     with pytest.raises(TypeError):
         ops.reduce_by_key_sorted(keys, None, op="min")
