@@ -23,11 +23,64 @@ def _bucket_to_rank(world: int, device) -> torch.Tensor:
     return (b * world) >> 8
 
 
+SAMPLE_OVER = 64  # splitter samples per rank per destination
+
+
 class TeraSortJob:
-    def __init__(self, device, group=None):
+    """partitioner:
+      "topbyte" (default) — one radix pass buckets by the key's top byte,
+        contiguous byte ranges per rank.  Zero extra sorts; right for
+        uniform keys (TeraGen's are).
+      "sample"  — sampling-based splitters: each rank contributes a
+        strided sample of its SORTED keys, every rank derives identical
+        quantile splitters from the all-gathered pool, and partitions are
+        searchsorted ranges.  Balances SKEWED key distributions (all keys
+        sharing a top byte land on one rank under "topbyte"); costs one
+        extra local sort + a small all-gather."""
+
+    def __init__(self, device, group=None, partitioner: str = "topbyte"):
         self.device = torch.device(device)
         self.group = group
         self.rank, self.world = dx.world_info(group)
+        assert partitioner in ("topbyte", "sample"), partitioner
+        self.partitioner = partitioner
+
+    def _sample_exchange(self, keys: torch.Tensor,
+                         payloads: Optional[torch.Tensor]):
+        """Sort-first splitter partitioning (the "sample" path)."""
+        import torch.distributed as td
+
+        sk, sv = ops.sort_pairs(keys, payloads, bits=64)
+        n = sk.numel()
+        S = SAMPLE_OVER * self.world
+        # fixed-size strided sample of the sorted keys (u64 order);
+        # empty ranks contribute +max keys so they sort to the end
+        if n:
+            idx = torch.tensor([min(n - 1, n * i // S) for i in range(S)],
+                               dtype=torch.int64, device=sk.device)
+            samp = sk.index_select(0, idx)
+        else:
+            samp = torch.full((S,), -1, dtype=torch.int64, device=sk.device)
+        pool = [torch.empty_like(samp) for _ in range(self.world)]
+        td.all_gather(pool, samp.contiguous(), group=self.group)
+        # identical on every rank: sort the pool in u64 order, take the
+        # world-1 quantile cuts as splitters
+        pool_s = torch.cat(pool) ^ (-1 << 63)
+        pool_s, _ = torch.sort(pool_s)
+        cuts = torch.tensor(
+            [pool_s.numel() * j // self.world for j in range(1, self.world)],
+            dtype=torch.int64, device=pool_s.device)
+        splitters = pool_s.index_select(0, cuts)
+        # partition boundaries in the sorted local array (u64 order)
+        bnd = torch.searchsorted(sk ^ (-1 << 63), splitters)
+        bl = [0] + bnd.cpu().tolist() + [n]
+        send = [bl[i + 1] - bl[i] for i in range(self.world)]
+        send_t = torch.tensor(send, dtype=torch.int64, device=sk.device)
+        recv = dx.exchange_counts(send_t, self.group).cpu().tolist()
+        rkeys = dx.exchange(sk, send, recv, self.group)
+        rpl = (dx.exchange(sv, send, recv, self.group)
+               if payloads is not None else None)
+        return rkeys, rpl
 
     def run(self, keys: torch.Tensor,
             payloads: Optional[torch.Tensor] = None
@@ -35,7 +88,9 @@ class TeraSortJob:
         """keys: i64 (u64 bit order); returns this rank's globally-ordered
         shard (rank-major partitioning over the sorted key space)."""
         dev = self.device
-        if self.world > 1:
+        if self.world > 1 and self.partitioner == "sample":
+            keys, payloads = self._sample_exchange(keys, payloads)
+        elif self.world > 1:
             if dev.type == "cuda":
                 pl = payloads if payloads is not None else torch.empty(
                     0, dtype=torch.int64, device=dev)

@@ -148,3 +148,71 @@ def _ii_worker(rank, world, port):
 def test_inverted_index_gloo_ws2():
     torch.multiprocessing.spawn(_ii_worker, args=(2, _free_port()),
                                 nprocs=2, join=True)
+
+
+def _ts_sample_worker(rank, world, port):
+    """Skewed keys (all top bytes 0) — "topbyte" would send everything to
+    rank 0; sampled splitters balance the ranks and preserve global
+    order (rank-major by splitter range)."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.distributed.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        rng = np.random.default_rng(90 + rank)
+        n = 20_000
+        keys = rng.integers(0, 2 ** 40, size=n, dtype=np.uint64)  # top byte 0
+        pl = rng.integers(0, 2 ** 62, size=n, dtype=np.uint64)
+        job = TeraSortJob("cpu", partitioner="sample")
+        sk, sv = job.run(torch.from_numpy(keys.view(np.int64)),
+                         torch.from_numpy(pl.view(np.int64)))
+        assert job.validate(sk)
+        mine = _u64(sk)
+        # balance: no rank holds more than 65% of the total
+        sizes = [None] * world
+        torch.distributed.all_gather_object(sizes, len(mine))
+        assert max(sizes) <= 0.65 * sum(sizes), sizes
+        # payloads still ride with their keys (post-exchange a rank holds
+        # keys from EVERY rank — pair against the gathered global map)
+        pairs = [None] * world
+        torch.distributed.all_gather_object(
+            pairs, list(zip(keys.tolist(), pl.tolist())))
+        kp = {k: v for plist in pairs for k, v in plist}
+        pv = _u64(sv)
+        for i in range(0, len(mine), max(1, len(mine) // 64)):
+            assert kp[int(mine[i])] == int(pv[i])
+        # global order: every key on rank r sorts <= every key on rank r+1
+        ends = [None] * world
+        torch.distributed.all_gather_object(
+            ends, (int(mine[0]), int(mine[-1])) if len(mine) else None)
+        all_keys = [None] * world
+        torch.distributed.all_gather_object(all_keys, mine.tolist())
+        all_in = [None] * world
+        torch.distributed.all_gather_object(all_in, keys.tolist())
+        if rank == 0:
+            prev_end = None
+            for e in ends:
+                if e is None:
+                    continue
+                if prev_end is not None:
+                    assert prev_end <= e[0]
+                prev_end = e[1]
+            got = sorted(x for l in all_keys for x in l)
+            exp = sorted(x for l in all_in for x in l)
+            assert got == exp
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_terasort_sampled_splitters_gloo_ws2():
+    torch.multiprocessing.spawn(_ts_sample_worker, args=(2, _free_port()),
+                                nprocs=2, join=True)
+
+
+def test_terasort_sample_single_rank():
+    rng = np.random.default_rng(4)
+    keys = rng.integers(0, 2 ** 30, size=5_000, dtype=np.uint64)
+    job = TeraSortJob("cpu", partitioner="sample")
+    sk, _ = job.run(torch.from_numpy(keys.view(np.int64)), None)
+    assert job.validate(sk)
+    assert np.array_equal(_u64(sk), np.sort(keys))
