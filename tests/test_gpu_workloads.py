@@ -64,3 +64,19 @@ def test_inverted_index_gpu(dev):
     for w in (vocab[0], vocab[123], vocab[499]):
         assert res.lookup(w) == exp[w]
     assert res.lookup(b"no-such-word-xyz") == []
+
+
+def test_terasort_sample_partitioner_gpu(dev):
+    """Single-rank "sample" path on hardware (the collectives are covered
+    by the gloo ws=2 test; here the sort-first code runs on the HIP
+    radix sort)."""
+    import numpy as np
+    from mapreduce_amd.gpu.terasort import TeraSortJob
+    rng = np.random.default_rng(21)
+    keys_np = rng.integers(0, 2 ** 40, size=2_000_000, dtype=np.uint64)
+    keys = torch.from_numpy(keys_np.view(np.int64)).to(dev)
+    job = TeraSortJob(dev, partitioner="sample")
+    sk, _ = job.run(keys, None)
+    assert job.validate(sk)
+    assert np.array_equal(sk.cpu().numpy().view(np.uint64),
+                          np.sort(keys_np))
