@@ -1,0 +1,71 @@
+"""Per-key extremes (min/max) — the canonical *idempotent* reducer.
+
+Task-script shape per the reference contract (SURVEY.md §2.3;
+examples/WordCount/init.lua:55-64): one module provides every role.
+mapfn emits (station, (t, t)) and the reducer folds (lo, hi) envelopes,
+which is associative+commutative+idempotent — so the engine may combine
+eagerly and re-apply safely (job.lua:104-106, 264-274).  The GPU tier's
+segmented min/max kernels (`ops.reduce_by_key_sorted(op="min"|"max")`)
+run this same reduction on HBM-resident columns.
+
+init({"readings": {station: [temps]}}) or {"nstations": N, "n": M} for
+synthetic data.
+"""
+
+from __future__ import annotations
+
+import random
+
+CONF: dict = {}
+RESULTS: dict = {}
+
+associative_reducer = True
+commutative_reducer = True
+idempotent_reducer = True
+
+
+def init(args):
+    CONF.clear()
+    RESULTS.clear()
+    args = args or {}
+    if "readings" in args:
+        CONF["readings"] = dict(args["readings"])
+    else:
+        rng = random.Random(args.get("seed", 0))
+        ns = int(args.get("nstations", 8))
+        n = int(args.get("n", 200))
+        CONF["readings"] = {
+            f"s{i:02d}": [rng.uniform(-40.0, 45.0) for _ in range(n)]
+            for i in range(ns)
+        }
+
+
+def taskfn(emit):
+    # one map job per station (split = the station's reading list)
+    for station in sorted(CONF["readings"]):
+        emit(station, station)
+
+
+def mapfn(key, value, emit):
+    for t in CONF["readings"][value]:
+        emit(value, (t, t))
+
+
+def partitionfn(key):
+    return sum(key.encode()) % 4
+
+
+def combinerfn(key, values, emit):
+    lo = min(v[0] for v in values)
+    hi = max(v[1] for v in values)
+    emit((lo, hi))
+
+
+reducefn = combinerfn
+
+
+def finalfn(pairs):
+    for station, values in pairs:
+        (lo, hi), = values
+        RESULTS[station] = (lo, hi)
+    return True

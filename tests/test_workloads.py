@@ -216,3 +216,18 @@ def test_terasort_sample_single_rank():
     sk, _ = job.run(torch.from_numpy(keys.view(np.int64)), None)
     assert job.validate(sk)
     assert np.array_equal(_u64(sk), np.sort(keys))
+
+
+def test_extremes_example():
+    """Per-key min/max task script (idempotent reducer) vs a direct
+    oracle; the same module serves every role (INIT-SCRIPT case)."""
+    from mapreduce_amd import run_local
+    from mapreduce_amd.examples import extremes
+
+    extremes.init({"nstations": 6, "n": 100, "seed": 3})
+    srv = run_local({"fns": {r: extremes for r in (
+        "taskfn", "mapfn", "partitionfn", "reducefn", "combinerfn",
+        "finalfn")}, "verbose": False}, nworkers=2)
+    assert srv.finished
+    exp = {s: (min(v), max(v)) for s, v in extremes.CONF["readings"].items()}
+    assert extremes.RESULTS == exp
