@@ -216,3 +216,52 @@ def test_inverted_index_pair_iterator():
     got = dict(res.pair_iterator(order="lex"))
     assert got == exp
     assert list(dict(res.pair_iterator(order="lex"))) == sorted(exp)
+
+
+def _empty_rank_worker(rank, world, port):
+    """One rank has an empty corpus; chunked rounds still agree and the
+    non-empty rank's counts survive (zero-send boundary paths)."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.distributed.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        os.environ["MR_SHUFFLE_BUDGET_BYTES"] = "2048"
+        if rank == 0:
+            c = make_corpus("cpu", nwords=4_000, nsplits=2, vocab_size=300,
+                            seed=21)
+            text, splits = c.text, c.splits()
+        else:
+            text = torch.zeros(0, dtype=torch.uint8)
+            splits = [(0, 0)]
+        job = WordCountJob("cpu", vocab_estimate=600)
+        res = job.run(text, splits)
+        all_pairs = [None] * world
+        torch.distributed.all_gather_object(all_pairs, res.to_host())
+        all_n = [None] * world
+        torch.distributed.all_gather_object(all_n, res.nwords)
+        assert all_n == [4_000, 0]
+        all_texts = [None] * world
+        torch.distributed.all_gather_object(
+            all_texts, bytes(text.numpy().tobytes()))
+        if rank == 0:
+            got = collections.Counter()
+            for plist in all_pairs:
+                for w, n in plist:
+                    got[w] += n
+            exp = collections.Counter()
+            for t in all_texts:
+                exp.update(t.split())
+            assert got == exp
+    finally:
+        os.environ.pop("MR_SHUFFLE_BUDGET_BYTES", None)
+        torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_chunked_shuffle_empty_rank_gloo_ws2():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    torch.multiprocessing.spawn(_empty_rank_worker, args=(2, port), nprocs=2,
+                                join=True)
