@@ -255,3 +255,25 @@ def test_restore_reduce_phase_skips_map():
     assert srv.finished
     assert executed_maps == []  # map skipped
     assert results == {"k0": 3, "k1": 3}
+
+
+def test_worker_idle_backoff_sequence(monkeypatch):
+    """Idle backoff grows x1.5 per empty iteration and caps at max_sleep
+    (worker.lua:100-101); a LocalCoordinator with no task keeps every
+    iteration idle, so the recorded sleeps are the raw schedule."""
+    from mapreduce_amd.parallel.coord import LocalCoordinator
+    from mapreduce_amd.worker import Worker
+    import mapreduce_amd.worker as worker_mod
+
+    slept = []
+    monkeypatch.setattr(worker_mod.time, "sleep", slept.append)
+    w = Worker(coord=LocalCoordinator()).configure(
+        {"max_iter": 8, "max_sleep": 3.0, "min_sleep": 1.0})
+    w._worker_execute()
+    exp = []
+    s = 1.0
+    for _ in range(8):
+        exp.append(s)
+        s = min(s * 1.5, 3.0)
+    assert slept == pytest.approx(exp)  # 1, 1.5, 2.25, 3, 3, ...
+    assert max(slept) == 3.0
