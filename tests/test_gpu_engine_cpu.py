@@ -265,3 +265,56 @@ def test_chunked_shuffle_empty_rank_gloo_ws2():
     s.close()
     torch.multiprocessing.spawn(_empty_rank_worker, args=(2, port), nprocs=2,
                                 join=True)
+
+
+def test_pipelined_equals_sequential():
+    """Depth-2 pipelined driver returns bit-identical results to
+    sequential runs (CPU tier degrades to exact sequential execution;
+    on GPU only the issue order differs)."""
+    from mapreduce_amd.gpu.pipeline import PipelinedWordCount
+
+    c = make_corpus("cpu", nwords=8_000, nsplits=4, vocab_size=400, seed=5)
+    seq = WordCountJob("cpu", vocab_estimate=800)
+    ref = seq.run(c.text, c.splits())
+    ref_pairs = sorted(ref.to_host())
+    for use_runner in (False, True):
+        pipe = PipelinedWordCount("cpu", vocab_estimate=800,
+                                  use_runner=use_runner)
+        for _ in range(3):
+            res = pipe.step(c.text, c.splits())
+            assert res.nwords == ref.nwords
+            assert sorted(res.to_host()) == ref_pairs
+        tail = pipe.flush()
+        assert sorted(tail.to_host()) == ref_pairs
+
+
+def _pipe_dist_worker(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.distributed.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from mapreduce_amd.gpu.pipeline import PipelinedWordCount
+
+        c = make_corpus("cpu", nwords=6_000, nsplits=3, vocab_size=300,
+                        seed=33 + rank)
+        seq = WordCountJob("cpu", vocab_estimate=600)
+        ref_pairs = sorted(seq.run(c.text, c.splits()).to_host())
+        pipe = PipelinedWordCount("cpu", vocab_estimate=600, use_runner=True)
+        for _ in range(3):
+            res = pipe.step(c.text, c.splits())
+            assert sorted(res.to_host()) == ref_pairs
+        pipe.flush()
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_pipelined_gloo_ws2():
+    """Collective order under lookahead: both ranks run the same
+    schedule, so the interleaved barriers/all-to-alls pair up."""
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    torch.multiprocessing.spawn(_pipe_dist_worker, args=(2, port), nprocs=2,
+                                join=True)
