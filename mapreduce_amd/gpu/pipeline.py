@@ -54,14 +54,13 @@ class PipelinedWordCount:
         if use_runner:
             from .runner import GpuClusterRunner
 
-            self.runners = [GpuClusterRunner(j, claim_mode=claim_mode)
-                            for j in self.jobs]
-            # both engine instances share one control plane; namespace the
-            # job docs per instance so in-flight docs never collide
-            for i, r in enumerate(self.runners):
-                r.worker_name = f"rank{r.rank}p{i}"
-                ns = r._ns
-                r._ns = (lambda ns=ns, i=i: f"{ns()}_p{i}")
+            # both engine instances share one control plane; the suffix
+            # namespaces the job docs so in-flight docs never collide
+            self.runners = [
+                GpuClusterRunner(j, claim_mode=claim_mode,
+                                 ns_suffix=f"_p{i}")
+                for i, j in enumerate(self.jobs)
+            ]
         self.cur = 0
         self._inflight = False
 

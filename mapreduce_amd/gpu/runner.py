@@ -43,18 +43,23 @@ def default_coordinator(world: int) -> Coordinator:
 
 class GpuClusterRunner:
     def __init__(self, job, coord: Optional[Coordinator] = None,
-                 group=None, claim_mode: str = "batch"):
+                 group=None, claim_mode: str = "batch",
+                 ns_suffix: str = ""):
+        """ns_suffix namespaces this runner's job docs — several runners
+        (e.g. the two pipelined engine instances) can share one control
+        plane without their in-flight docs colliding."""
         self.job = job
         self.rank, self.world = dx.world_info(group)
         self.group = group
         self.coord = coord or default_coordinator(self.world)
         self.task = Task(self.coord)
         self.claim_mode = claim_mode
-        self.worker_name = f"rank{self.rank}"
+        self.ns_suffix = ns_suffix
+        self.worker_name = f"rank{self.rank}{ns_suffix}"
 
     # ------------------------------------------------------------- phases
     def _ns(self) -> str:
-        return f"{Task.MAP_JOBS}_r{self.rank}"
+        return f"{Task.MAP_JOBS}_r{self.rank}{self.ns_suffix}"
 
     def _insert_map_jobs(self, splits: List[Tuple[int, int]]):
         ns = self._ns()

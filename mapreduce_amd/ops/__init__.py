@@ -202,6 +202,8 @@ def reduce_by_key_sorted(keys: torch.Tensor,
     if op != "sum" and (vals is None or
                         vals.dtype not in (torch.int64, torch.float64)):
         raise TypeError("min/max reduction needs i64 or f64 vals")
+    if op != "sum" and vals.numel() != keys.numel():
+        raise ValueError("min/max reduction needs one value per key")
     if not keys.is_cuda:
         from . import _cpu
         return _cpu.reduce_by_key_sorted(keys, vals, aux, op)
