@@ -67,6 +67,15 @@ def main() -> int:
     # per-phase events of a single instance) restores sequential steps.
     import os
     use_pipe = os.environ.get("MR_PIPELINE", "1") != "0" and not args.timing
+    if use_pipe and device.type == "cuda":
+        # two engine instances double the spill allocation (~8 bytes per
+        # corpus byte each); at giant corpora that exceeds HBM — fall
+        # back to sequential (measured: 64x Europarl = 19.6 GB corpus,
+        # 2 x 157 GB spill > 288 GB)
+        n = corpus.text.numel()
+        total = torch.cuda.get_device_properties(device).total_memory
+        if n + 2 * 8.2 * n > 0.85 * total:
+            use_pipe = False
     job = WordCountJob(device, vocab_estimate=vocab_est,
                        mode=args.mode, timing=args.timing)
     pipe = None
