@@ -80,3 +80,23 @@ def test_terasort_sample_partitioner_gpu(dev):
     assert job.validate(sk)
     assert np.array_equal(sk.cpu().numpy().view(np.uint64),
                           np.sort(keys_np))
+
+
+def test_pipelined_wordcount_gpu(dev):
+    """Depth-2 two-stream pipeline returns results identical to the
+    sequential engine on hardware (the bench's default driver)."""
+    from mapreduce_amd.gpu.corpus import make_corpus
+    from mapreduce_amd.gpu.pipeline import PipelinedWordCount
+    from mapreduce_amd.gpu.wordcount import WordCountJob
+
+    c = make_corpus(dev, nwords=300_000, nsplits=8, vocab_size=5_000,
+                    seed=17)
+    ref = WordCountJob(dev, vocab_estimate=8_000).run(c.text, c.splits())
+    ref_pairs = sorted(ref.to_host())
+    pipe = PipelinedWordCount(dev, vocab_estimate=8_000, use_runner=True)
+    for _ in range(3):
+        res = pipe.step(c.text, c.splits())
+        assert res.nwords == ref.nwords
+        assert sorted(res.to_host()) == ref_pairs
+    tail = pipe.flush()
+    assert sorted(tail.to_host()) == ref_pairs
