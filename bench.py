@@ -11,8 +11,12 @@ per GPU; weak scaling: each rank owns one Europarl-size corpus).
 One step = one complete MapReduce job over the corpus: fused
 tokenize+combine kernels, unique extraction + radix sort, RCCL all-to-all
 shuffle, sort+segmented reduce, per-rank sorted results materialized
-(counts ready for the finalfn boundary).  Reference headline to beat:
-49.23 s / ~1.0 M words/s on 4 CPU workers.
+(counts ready for the finalfn boundary).  Back-to-back steps run under
+the depth-2 two-stream job pipeline (gpu/pipeline.py) by default — every
+job still executes completely and exactly one job's work falls in each
+timed step (steady state); MR_PIPELINE=0 forces sequential steps.
+Reference headline to beat: 49.23 s / ~1.0 M words/s on 4 CPU workers
+(BASELINE.md).
 """
 
 from __future__ import annotations
