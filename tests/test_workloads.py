@@ -231,3 +231,76 @@ def test_extremes_example():
     assert srv.finished
     exp = {s: (min(v), max(v)) for s, v in extremes.CONF["readings"].items()}
     assert extremes.RESULTS == exp
+
+
+def _keyed_oracle(keys, vals, op):
+    import numpy as np
+    agg = {}
+    f = {"sum": lambda a, b: a + b, "min": min, "max": max}[op]
+    for k, v in zip(keys.tolist(), vals.tolist()):
+        agg[k] = f(agg[k], v) if k in agg else v
+    uk = np.sort(np.array(list(agg), dtype=np.uint64))
+    return uk, np.array([agg[int(k)] for k in uk])
+
+
+def test_keyed_reduce_single_rank():
+    from mapreduce_amd.gpu.keyed_reduce import KeyedReduceJob
+
+    rng = np.random.default_rng(8)
+    keys = rng.integers(0, 500, size=30_000, dtype=np.uint64) * 7919
+    for op in ("sum", "min", "max"):
+        for vals_np in (rng.integers(-10 ** 9, 10 ** 9, size=30_000,
+                                     dtype=np.int64),
+                        rng.standard_normal(30_000)):
+            job = KeyedReduceJob("cpu", op=op)
+            uk, uv = job.run(torch.from_numpy(keys.view(np.int64)),
+                             torch.from_numpy(vals_np))
+            ek, ev = _keyed_oracle(keys, vals_np, op)
+            assert np.array_equal(_u64(uk), ek)
+            if vals_np.dtype == np.int64:
+                assert np.array_equal(uv.numpy(), ev)
+            else:
+                assert np.allclose(uv.numpy(), ev, rtol=1e-12, atol=1e-9)
+
+
+def _kr_worker(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.distributed.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from mapreduce_amd.gpu.keyed_reduce import KeyedReduceJob
+
+        rng = np.random.default_rng(60 + rank)
+        keys = rng.integers(0, 300, size=10_000, dtype=np.uint64) * 104729
+        vals = rng.integers(-10 ** 6, 10 ** 6, size=10_000, dtype=np.int64)
+        job = KeyedReduceJob("cpu", op="min")
+        uk, uv = job.run(torch.from_numpy(keys.view(np.int64)),
+                         torch.from_numpy(vals))
+        # ownership: mulhi partition
+        parts = ((_u64(uk).astype(object) * world) >> 64).astype(int)
+        assert (parts == rank).all()
+        all_pairs = [None] * world
+        torch.distributed.all_gather_object(
+            all_pairs, (_u64(uk).tolist(), uv.tolist()))
+        all_in = [None] * world
+        torch.distributed.all_gather_object(
+            all_in, (keys.tolist(), vals.tolist()))
+        if rank == 0:
+            got = {}
+            for ks, vs in all_pairs:
+                for k, v in zip(ks, vs):
+                    assert k not in got, "key owned by two ranks"
+                    got[k] = v
+            exp = {}
+            for ks, vs in all_in:
+                for k, v in zip(ks, vs):
+                    exp[k] = min(exp.get(k, v), v)
+            assert got == exp
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_keyed_reduce_gloo_ws2():
+    torch.multiprocessing.spawn(_kr_worker, args=(2, _free_port()),
+                                nprocs=2, join=True)
