@@ -100,3 +100,27 @@ def test_pipelined_wordcount_gpu(dev):
         assert sorted(res.to_host()) == ref_pairs
     tail = pipe.flush()
     assert sorted(tail.to_host()) == ref_pairs
+
+
+def test_keyed_reduce_gpu(dev):
+    """KeyedReduceJob on hardware (composes the GPU-validated sort +
+    segmented min/max kernels; multi-rank covered by the gloo test)."""
+    import numpy as np
+    from mapreduce_amd.gpu.keyed_reduce import KeyedReduceJob
+
+    rng = np.random.default_rng(19)
+    keys_np = rng.integers(0, 2_000, size=1_000_000, dtype=np.uint64) * 7919
+    vals_np = rng.integers(-10 ** 12, 10 ** 12, size=1_000_000,
+                           dtype=np.int64)
+    keys = torch.from_numpy(keys_np.view(np.int64)).to(dev)
+    vals = torch.from_numpy(vals_np).to(dev)
+    exp_keys, idx = np.unique(keys_np, return_index=True)
+    order = np.argsort(keys_np, kind="stable")
+    sorted_vals = vals_np[order]
+    bounds = np.searchsorted(keys_np[order], exp_keys)
+    for op, red in (("sum", np.add), ("min", np.minimum),
+                    ("max", np.maximum)):
+        uk, uv = KeyedReduceJob(dev, op=op).run(keys, vals)
+        assert np.array_equal(uk.cpu().numpy().view(np.uint64), exp_keys)
+        assert np.array_equal(uv.cpu().numpy(),
+                              red.reduceat(sorted_vals, bounds))
