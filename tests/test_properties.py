@@ -136,3 +136,43 @@ def test_reduce_by_key_min_max_cpu():
         ops.reduce_by_key_sorted(keys, None, op="min")
     with pytest.raises(ValueError):
         ops.reduce_by_key_sorted(keys, vals, op="mean")
+
+
+@given(st.lists(st.integers(min_value=0, max_value=40), min_size=2,
+                max_size=5),
+       st.integers(min_value=1, max_value=9),
+       st.integers(min_value=0, max_value=2**32))
+@settings(max_examples=120, deadline=None)
+def test_chunked_shuffle_slicing_partitions_exactly(counts, rounds, seed):
+    """The chunked shuffle's round-slicing formulas (gpu/wordcount.py
+    _chunked_shuffle_reduce): for any per-partition counts and any round
+    count, the per-round element ranges partition each partition's
+    segment exactly, receivers derive the same per-round counts from the
+    totals alone, and the blob byte boundaries partition the bytes."""
+    import numpy as np
+
+    rng = np.random.default_rng(seed % (2**31))
+    world = len(counts)
+    n = sum(counts)
+    lens = rng.integers(1, 17, size=n, dtype=np.int64)
+    lcs = np.cumsum(lens)
+    ecs = np.concatenate([lcs - lens, [int(lens.sum())]])
+    poff = [0]
+    for c in counts:
+        poff.append(poff[-1] + c)
+    ebnd = [[poff[p] + counts[p] * r // rounds for r in range(rounds + 1)]
+            for p in range(world)]
+    for p in range(world):
+        # monotone, exact partition of [poff[p], poff[p+1])
+        assert ebnd[p][0] == poff[p] and ebnd[p][-1] == poff[p + 1]
+        assert all(ebnd[p][r] <= ebnd[p][r + 1] for r in range(rounds))
+        # receiver-side derivation from the total alone matches
+        for r in range(rounds):
+            sender = ebnd[p][r + 1] - ebnd[p][r]
+            receiver = (counts[p] * (r + 1) // rounds
+                        - counts[p] * r // rounds)
+            assert sender == receiver
+        # blob byte ranges partition the partition's bytes
+        tot = ecs[ebnd[p][-1]] - ecs[ebnd[p][0]]
+        assert sum(ecs[ebnd[p][r + 1]] - ecs[ebnd[p][r]]
+                   for r in range(rounds)) == tot
