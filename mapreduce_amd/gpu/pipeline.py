@@ -20,7 +20,7 @@ deterministic because every rank runs the same lookahead schedule.
 from __future__ import annotations
 
 import contextlib
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 import torch
 
